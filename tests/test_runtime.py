@@ -1,0 +1,129 @@
+"""Actor runtime: spawn, RPC, meshes, serialization, teardown."""
+
+import asyncio
+
+import pytest
+import torch
+
+from torchstore_amd.runtime import (
+    Actor,
+    ActorMesh,
+    close_connections,
+    endpoint,
+    spawn_actors,
+    actor_context,
+)
+from torchstore_amd.runtime import serde
+
+
+class Echo(Actor):
+    def __init__(self, tag="t"):
+        self.tag = tag
+        self.items = {}
+
+    @endpoint
+    def ping(self):
+        return f"pong-{self.tag}-{actor_context().rank}"
+
+    @endpoint
+    def add(self, a, b):
+        return a + b
+
+    @endpoint
+    async def put(self, key, tensor):
+        self.items[key] = tensor
+        return tensor.shape
+
+    @endpoint
+    async def get(self, key):
+        return self.items[key]
+
+    @endpoint
+    def boom(self):
+        raise KeyError("missing-thing")
+
+    @endpoint
+    def my_rank(self):
+        return actor_context().rank
+
+
+def test_serde_roundtrip_tensor():
+    t = torch.randn(33, 7)
+    header, bufs = serde.dumps({"x": t, "n": 5})
+    raw = [bytearray(b) for b in bufs]
+    out = serde.loads(header, raw)
+    assert out["n"] == 5
+    assert torch.equal(out["x"], t)
+
+
+def test_serde_bf16_and_empty():
+    t = torch.randn(8, dtype=torch.float32).to(torch.bfloat16)
+    e = torch.empty(0, 4)
+    header, bufs = serde.dumps((t, e))
+    out = serde.loads(header, [bytearray(b) for b in bufs])
+    assert torch.equal(out[0], t)
+    assert out[1].shape == (0, 4)
+
+
+class _StrippingBuf:
+    def __init__(self):
+        self.local = torch.randn(4)
+        self.meta = "m"
+
+    def __getstate__(self):
+        d = self.__dict__.copy()
+        d["local"] = None
+        return d
+
+
+def test_serde_getstate_strips():
+    header, bufs = serde.dumps(_StrippingBuf())
+    assert len(bufs) == 0  # the tensor was stripped before pickling
+    out = serde.loads(header, [])
+    assert out.local is None and out.meta == "m"
+
+
+async def test_spawn_call_roundtrip():
+    mesh = spawn_actors(2, Echo, "echo-test", tag="hello")
+    try:
+        results = await mesh.ping.call()
+        assert results == ["pong-hello-0", "pong-hello-1"]
+        one = await mesh.handles[1].add.call_one(2, 3)
+        assert one == 5
+        ranks = await mesh.my_rank.call()
+        assert ranks == [0, 1]
+    finally:
+        await mesh.stop()
+        await close_connections()
+
+
+async def test_tensor_rpc_and_exceptions():
+    mesh = spawn_actors(1, Echo, "echo-t2")
+    try:
+        h = mesh.handles[0]
+        t = torch.randn(128, 64)
+        shape = await h.put.call_one("k", t)
+        assert tuple(shape) == (128, 64)
+        back = await h.get.call_one("k")
+        assert torch.equal(back, t)
+        with pytest.raises(KeyError):
+            await h.boom.call_one()
+        # non-endpoint methods are rejected
+        with pytest.raises(AttributeError):
+            await h.nonexistent.call_one()
+    finally:
+        await mesh.stop()
+        await close_connections()
+
+
+async def test_mesh_slice_and_concurrent_calls():
+    mesh = spawn_actors(4, Echo, "echo-m", mesh_shape=(2, 2))
+    try:
+        assert mesh.slice((1, 0)).rank == 2
+        outs = await asyncio.gather(
+            *(mesh.handles[i % 4].add.call_one(i, i) for i in range(32))
+        )
+        assert outs == [2 * i for i in range(32)]
+    finally:
+        await mesh.stop()
+        await close_connections()

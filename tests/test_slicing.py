@@ -1,0 +1,128 @@
+"""Slice math: intersection, region extraction, assembly, byte views."""
+
+import pytest
+import torch
+
+from torchstore_amd.ops.slicing import (
+    assemble,
+    bounding_box,
+    byte_view,
+    extract_region,
+    is_dense_view,
+    overlap,
+    region_view,
+    same_memory,
+)
+from torchstore_amd.types import TensorSlice
+
+
+def test_overlap_basic():
+    assert overlap((0, 0), (4, 4), (2, 2), (4, 4)) == ((2, 2), (2, 2))
+    assert overlap((0,), (4,), (4,), (4,)) is None
+    assert overlap((0, 0), (4, 4), (0, 4), (4, 4)) is None
+    # containment
+    assert overlap((0, 0), (8, 8), (2, 3), (2, 2)) == ((2, 3), (2, 2))
+
+
+def test_overlap_rank_mismatch():
+    with pytest.raises(ValueError):
+        overlap((0,), (1,), (0, 0), (1, 1))
+
+
+def test_extract_region_matches_torch_slicing():
+    t = torch.arange(64).reshape(8, 8)
+    # shard with global offset (4, 0) — its rows are global rows 4..12
+    shard_off = (4, 0)
+    view = extract_region(t, shard_off, (6, 2), (2, 3))
+    assert torch.equal(view, t[2:4, 2:5])
+
+
+def test_extract_region_out_of_bounds():
+    t = torch.zeros(4, 4)
+    with pytest.raises(IndexError):
+        extract_region(t, (0, 0), (2, 2), (4, 4))
+
+
+def test_region_view_writeback():
+    dest = torch.zeros(4, 4)
+    v = region_view(dest, (0, 0), (1, 1), (2, 2))
+    v.fill_(7.0)
+    assert dest[1:3, 1:3].eq(7).all()
+    assert not is_dense_view(v)
+    full_rows = region_view(dest, (0, 0), (2, 0), (2, 4))
+    assert is_dense_view(full_rows)
+
+
+def test_bounding_box():
+    parts = [((0, 0), (2, 4)), ((2, 0), (2, 4))]
+    assert bounding_box(parts) == ((0, 0), (4, 4))
+
+
+def test_assemble_tiling():
+    full = torch.arange(32, dtype=torch.float32).reshape(4, 8)
+    parts = [
+        ((0, 0), full[:2].clone()),
+        ((2, 0), full[2:].clone()),
+    ]
+    out, origin = assemble(parts)
+    assert origin == (0, 0)
+    assert torch.equal(out, full)
+
+
+def test_assemble_columns_and_offset_origin():
+    full = torch.randn(4, 8)
+    parts = [
+        ((2, 0), full[2:, :4].clone()),
+        ((2, 4), full[2:, 4:].clone()),
+    ]
+    out, origin = assemble(parts)
+    assert origin == (2, 0)
+    assert torch.equal(out, full[2:])
+
+
+def test_assemble_into_out():
+    full = torch.randn(4, 4)
+    dest = torch.zeros(4, 4)
+    assemble(
+        [((0, 0), full[:2].clone()), ((2, 0), full[2:].clone())],
+        out=dest,
+        out_offsets=(0, 0),
+    )
+    assert torch.equal(dest, full)
+
+
+def test_assemble_incomplete_raises():
+    with pytest.raises(ValueError):
+        assemble([((0, 0), torch.zeros(1, 4)), ((3, 0), torch.zeros(1, 4))])
+
+
+def test_byte_view():
+    t = torch.randn(5, 3, dtype=torch.float32)
+    b = byte_view(t)
+    assert b.dtype == torch.uint8 and b.numel() == 60
+    b[0] = 0xFF  # aliasing check: writes show through
+    assert t.view(torch.uint8).reshape(-1)[0] == 0xFF
+    bf = torch.randn(4).to(torch.bfloat16)
+    assert byte_view(bf).numel() == 8
+
+
+def test_same_memory():
+    t = torch.randn(8, 8)
+    assert same_memory(t, t[2:4])
+    assert same_memory(t, t[3, 1:2])
+    assert not same_memory(t, t.clone())
+
+
+def test_tensor_slice_intersect():
+    a = TensorSlice((0, 0), (4, 8), (8, 8), (0,), (2,))
+    b = TensorSlice((2, 0), (4, 8), (8, 8), (0,), (2,))
+    hit = a.intersect(b)
+    assert hit.offsets == (2, 0) and hit.local_shape == (2, 8)
+    c = TensorSlice((4, 0), (4, 8), (8, 8), (1,), (2,))
+    assert a.intersect(c) is None
+
+
+def test_expected_coordinates():
+    coords = set(TensorSlice.expected_coordinates((2, 3)))
+    assert len(coords) == 6
+    assert (1, 2) in coords
