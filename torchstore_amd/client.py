@@ -1,0 +1,342 @@
+"""LocalClient — data-plane orchestration in the user's process.
+
+Responsibilities (reference: torchstore ``client.py``):
+* build Requests from user values (DTensor → shard + TensorSlice);
+* pick the target volume per batch via the placement strategy;
+* expand a get into per-volume slice sub-requests (intersection of stored
+  vs requested shards), deduplicating replicated shards;
+* run all involved volumes' transports concurrently (asyncio.gather);
+* land bytes in-place when a destination tensor was provided, else
+  assemble fetched parts into the requested region;
+* keep the controller's index in sync (notify after put, before delete).
+"""
+
+from __future__ import annotations
+
+import asyncio
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import torch
+
+from torchstore_amd.controller import ObjectType, StorageInfo, VolumeInfo
+from torchstore_amd.ops.slicing import assemble, region_view, same_memory
+from torchstore_amd.runtime import ActorHandle
+from torchstore_amd.storage import OBJ_SENTINEL, TensorMeta
+from torchstore_amd.strategy import (
+    PlacementStrategy,
+    SingletonStrategy,
+    StorageVolumeRef,
+    strategy_from_spec,
+)
+from torchstore_amd.transport import TransportContext, create_transport
+from torchstore_amd.types import Request, TensorSlice
+from torchstore_amd.utils.logging import LatencyTracker, get_logger
+
+logger = get_logger("torchstore_amd.client")
+
+
+@dataclass
+class _SubFetch:
+    """One per-volume piece of a get."""
+
+    key: str
+    request: Request
+    # global region this piece covers (None for objects / whole tensors)
+    region_offsets: Optional[Tuple[int, ...]] = None
+    result: Any = None
+
+
+def _full_region_slice(global_shape: Sequence[int]) -> TensorSlice:
+    shape = tuple(global_shape)
+    return TensorSlice(
+        offsets=(0,) * len(shape),
+        local_shape=shape,
+        global_shape=shape,
+        coordinates=(),
+        mesh_shape=(),
+    )
+
+
+class LocalClient:
+    def __init__(
+        self,
+        controller: ActorHandle,
+        strategy: Optional[PlacementStrategy] = None,
+    ):
+        self._controller = controller
+        self._strategy = strategy
+        self._ctx = TransportContext()
+        self._volumes: Optional[Dict[str, VolumeInfo]] = None
+
+    # -- bring-up ---------------------------------------------------------
+    async def _ensure_volumes(self) -> Dict[str, VolumeInfo]:
+        if self._volumes is None:
+            infos, spec = await self._controller.get_volumes.call_one()
+            self._volumes = {v.volume_id: v for v in infos}
+            if self._strategy is None:
+                self._strategy = strategy_from_spec(spec)
+        return self._volumes
+
+    def _volume_ref(self, volume_id: str) -> StorageVolumeRef:
+        v = self._volumes[volume_id]
+        return StorageVolumeRef(
+            volume=v.handle,
+            volume_id=v.volume_id,
+            hostname=v.hostname,
+            device=v.device,
+            transport_context=self._ctx,
+            transport_type=self._strategy.transport if self._strategy else None,
+        )
+
+    # -- put --------------------------------------------------------------
+    async def put(self, key: str, value: Any) -> None:
+        await self.put_batch({key: value})
+
+    async def put_batch(self, items: Dict[str, Any]) -> None:
+        if not items:
+            return
+        volumes = await self._ensure_volumes()
+        requests = [Request.from_any(k, v) for k, v in items.items()]
+        volume_id = self._strategy.select_volume_id(list(volumes.keys()))
+        ref = self._volume_ref(volume_id)
+        tracker = LatencyTracker(f"put_batch[{len(requests)}]")
+        buffer = create_transport(ref)
+        await buffer.put(requests)
+        tracker.step("transport", sum(r.nbytes() for r in requests))
+        await self._controller.notify_put_batch.call_one(
+            [r.meta_only() for r in requests], volume_id
+        )
+        tracker.step("notify")
+
+    # -- get --------------------------------------------------------------
+    async def get(self, key: str, like: Any = None) -> Any:
+        result = await self.get_batch({key: like})
+        return result[key]
+
+    async def get_batch(self, fetches: Dict[str, Any]) -> Dict[str, Any]:
+        if not fetches:
+            return {}
+        await self._ensure_volumes()
+        located = await self._controller.locate.call_one(list(fetches.keys()))
+        plans = {
+            key: await self._plan_fetch(key, like, located[key])
+            for key, like in fetches.items()
+        }
+        # group sub-fetches per volume, fetch all volumes concurrently
+        by_volume: Dict[str, List[_SubFetch]] = {}
+        for subs in plans.values():
+            for volume_id, sf in subs:
+                by_volume.setdefault(volume_id, []).append(sf)
+        await asyncio.gather(
+            *(self._fetch_volume(vid, sfs) for vid, sfs in by_volume.items())
+        )
+        return {
+            key: self._finish_fetch(key, fetches[key], [sf for _, sf in subs])
+            for key, subs in plans.items()
+        }
+
+    async def _plan_fetch(
+        self, key: str, like: Any, locations: Dict[str, StorageInfo]
+    ) -> List[Tuple[str, _SubFetch]]:
+        """Decide which volumes to hit and what to ask each for."""
+        kinds = {info.object_type for info in locations.values()}
+        volume_ids = self._order_by_locality(locations.keys())
+
+        if kinds == {ObjectType.OBJECT}:
+            return [
+                (volume_ids[0], _SubFetch(key, Request(key=key, is_object=True)))
+            ]
+
+        dest, wanted = self._dest_and_region(like)
+
+        if ObjectType.TENSOR in kinds:
+            # at least one volume has the whole tensor — single-volume fetch
+            vid = next(
+                v for v in volume_ids
+                if locations[v].object_type == ObjectType.TENSOR
+            )
+            req = Request(key=key, tensor_slice=wanted, tensor_val=dest)
+            if dest is not None:
+                req.inplace = True
+            sf = _SubFetch(
+                key, req,
+                region_offsets=wanted.offsets if wanted is not None else None,
+            )
+            await self._allocate_dests(vid, [sf])
+            return [(vid, sf)]
+
+        # sharded key: expand into per-stored-shard intersections
+        if wanted is None:
+            # full-tensor fetch of a sharded key: region = whole global shape
+            any_slice = next(
+                iter(next(iter(locations.values())).tensor_slices)
+            )
+            wanted = _full_region_slice(any_slice.global_shape)
+
+        subs: List[Tuple[str, _SubFetch]] = []
+        covered: set = set()
+        for vid in volume_ids:
+            for stored in locations[vid].tensor_slices:
+                inter = stored.intersect(wanted)
+                if inter is None:
+                    continue
+                region = (inter.offsets, inter.local_shape)
+                if region in covered:
+                    continue  # replicated shard — fetch once
+                covered.add(region)
+                req = Request(key=key, tensor_slice=inter)
+                if dest is not None:
+                    view = region_view(
+                        dest, wanted.offsets, inter.offsets, inter.local_shape
+                    )
+                    req.tensor_val = view
+                    req.inplace = True
+                subs.append(
+                    (vid, _SubFetch(key, req, region_offsets=inter.offsets))
+                )
+        if not subs:
+            raise KeyError(
+                f"no stored shard of {key!r} overlaps the requested region"
+            )
+        for vid in {v for v, _ in subs}:
+            await self._allocate_dests(vid, [sf for v, sf in subs if v == vid])
+        return subs
+
+    def _order_by_locality(self, volume_ids) -> List[str]:
+        def sort_key(vid: str):
+            return (0 if self._volume_ref(vid).is_local else 1, vid)
+
+        return sorted(volume_ids, key=sort_key)
+
+    def _dest_and_region(
+        self, like: Any
+    ) -> Tuple[Optional[torch.Tensor], Optional[TensorSlice]]:
+        """Destination local tensor + wanted global region from a `like`."""
+        if like is None:
+            return None, None
+        from torch.distributed.tensor import DTensor
+
+        if isinstance(like, DTensor):
+            from torchstore_amd.types import (
+                _dtensor_is_trivially_local,
+                slice_from_dtensor,
+            )
+
+            if _dtensor_is_trivially_local(like):
+                local = like.to_local()
+                return local, _full_region_slice(local.shape)
+            return like.to_local(), slice_from_dtensor(like)
+        if isinstance(like, torch.Tensor):
+            return like, _full_region_slice(like.shape)
+        raise TypeError(f"cannot fetch into a {type(like)}")
+
+    async def _allocate_dests(self, volume_id: str, subs: List[_SubFetch]) -> None:
+        """Every tensor sub-request needs a destination before transport.get."""
+        missing = [
+            sf for sf in subs
+            if not sf.request.is_object and sf.request.tensor_val is None
+        ]
+        if not missing:
+            return
+        ref = self._volume_ref(volume_id)
+        metas = await ref.volume.get_meta.call_one(
+            [sf.request.meta_only() for sf in missing]
+        )
+        for sf, meta in zip(missing, metas):
+            if meta == OBJ_SENTINEL:
+                sf.request.is_object = True
+                continue
+            device = self._alloc_device(meta)
+            sf.request.tensor_val = torch.empty(
+                meta.shape, dtype=meta.dtype, device=device
+            )
+
+    @staticmethod
+    def _alloc_device(meta: TensorMeta) -> torch.device:
+        if meta.device == "cuda" and torch.cuda.is_available():
+            return torch.device("cuda", torch.cuda.current_device())
+        return torch.device("cpu")
+
+    async def _fetch_volume(self, volume_id: str, subs: List[_SubFetch]) -> None:
+        ref = self._volume_ref(volume_id)
+        buffer = create_transport(ref)
+        results = await buffer.get([sf.request for sf in subs])
+        for sf, res in zip(subs, results):
+            sf.result = res
+
+    def _finish_fetch(self, key: str, like: Any, subs: List[_SubFetch]) -> Any:
+        if len(subs) == 1 and subs[0].region_offsets is None:
+            sf = subs[0]
+            return self._return_like(like, sf.result)
+        dest, wanted = self._dest_and_region(like)
+        if dest is not None:
+            # verify every piece landed inside dest; copy any that didn't
+            for sf in subs:
+                if isinstance(sf.result, torch.Tensor) and not same_memory(
+                    dest, sf.result
+                ):
+                    view = region_view(
+                        dest, wanted.offsets, sf.region_offsets,
+                        tuple(sf.result.shape),
+                    )
+                    view.copy_(sf.result)
+            return like if like is not None else dest
+        parts = [
+            (sf.region_offsets, sf.result)
+            for sf in subs
+            if isinstance(sf.result, torch.Tensor)
+        ]
+        out, _origin = assemble(parts)
+        return out
+
+    def _return_like(self, like: Any, result: Any) -> Any:
+        if like is None:
+            return result
+        if isinstance(result, torch.Tensor):
+            dest, _ = self._dest_and_region(like)
+            if dest is not None and not same_memory(dest, result):
+                dest.copy_(result)
+            return like
+        return result
+
+    # -- delete / keys / exists ------------------------------------------
+    async def delete(self, key: str, missing_ok: bool = False) -> None:
+        await self._ensure_volumes()
+        volume_ids = await self._controller.notify_delete.call_one(
+            key, missing_ok
+        )
+        await asyncio.gather(
+            *(
+                self._volumes[vid].handle.delete.call_one(key, True)
+                for vid in volume_ids
+            )
+        )
+        self._ctx.drop_key(key)
+
+    async def delete_batch(self, keys: Sequence[str], missing_ok: bool = True) -> None:
+        await self._ensure_volumes()
+        removed = await self._controller.notify_delete_batch.call_one(
+            list(keys), missing_ok
+        )
+        by_volume: Dict[str, List[str]] = {}
+        for key, vids in removed.items():
+            for vid in vids:
+                by_volume.setdefault(vid, []).append(key)
+        await asyncio.gather(
+            *(
+                self._volumes[vid].handle.delete_batch.call_one(ks, True)
+                for vid, ks in by_volume.items()
+            )
+        )
+        for key in keys:
+            self._ctx.drop_key(key)
+
+    async def keys(self, prefix: Optional[str] = None) -> List[str]:
+        return await self._controller.list_keys.call_one(prefix)
+
+    async def exists(self, key: str) -> bool:
+        return await self._controller.key_exists.call_one(key)
+
+    def close(self) -> None:
+        self._ctx.close()

@@ -1,0 +1,129 @@
+"""state_dict exchange: flatten → batched put, with a commit marker.
+
+Reference semantics (torchstore ``state_dict_utils.py``):
+
+* ``put_state_dict`` flattens the nested dict (torch DCP's util), optionally
+  casts floating tensors to a transfer dtype, batch-puts every entry under
+  ``"{key}/{flat_key}"`` and writes ``"{key}/<MAPPING>"`` **last** — readers
+  fetch the mapping first, so a partially-pushed state_dict is invisible
+  (the commit-marker protocol);
+* ``get_state_dict`` fetches the mapping (missing ⇒ "no matching push"),
+  batch-gets all entries (in place into the user's state_dict tensors when
+  given — DTensor entries reshard per-entry), and unflattens.
+
+On GPU the dtype cast uses the fused cast+pack HIP kernel (K3) so the
+staging copy and the cast are one kernel pass.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, Optional
+
+import torch
+
+from torchstore_amd.client import LocalClient
+from torchstore_amd.utils.logging import LatencyTracker, get_logger
+
+logger = get_logger("torchstore_amd.state_dict")
+
+MAPPING_KEY = "<MAPPING>"
+
+
+def _flatten(state_dict: Dict[str, Any]):
+    from torch.distributed.checkpoint._nested_dict import flatten_state_dict
+
+    return flatten_state_dict(state_dict)
+
+
+def _unflatten(flat: Dict[str, Any], mapping):
+    from torch.distributed.checkpoint._nested_dict import unflatten_state_dict
+
+    return unflatten_state_dict(flat, mapping)
+
+
+def _cast_floating(
+    flat: Dict[str, Any], dtype: Optional[torch.dtype]
+) -> Dict[str, Any]:
+    if dtype is None:
+        return flat
+    from torchstore_amd.ops.cast import cast_tensor
+
+    out = {}
+    for k, v in flat.items():
+        if isinstance(v, torch.Tensor) and v.is_floating_point():
+            out[k] = cast_tensor(v, dtype)
+        else:
+            out[k] = v
+    return out
+
+
+def _nbytes(flat: Dict[str, Any]) -> int:
+    total = 0
+    for v in flat.values():
+        if isinstance(v, torch.Tensor):
+            total += v.numel() * v.element_size()
+    return total
+
+
+async def put_state_dict(
+    client: LocalClient,
+    state_dict: Dict[str, Any],
+    key: str,
+    transfer_dtype: Optional[torch.dtype] = None,
+) -> None:
+    tracker = LatencyTracker(f"put_state_dict[{key}]")
+    flat, mapping = _flatten(state_dict)
+    tracker.step("flatten")
+    flat = _cast_floating(flat, transfer_dtype)
+    tracker.step("cast")
+    await client.put_batch({f"{key}/{k}": v for k, v in flat.items()})
+    tracker.step("put_batch", _nbytes(flat))
+    # commit marker: written last, fetched first by readers
+    await client.put(f"{key}/{MAPPING_KEY}", mapping)
+    tracker.step("commit")
+    tracker.e2e(_nbytes(flat))
+
+
+async def get_state_dict(
+    client: LocalClient,
+    key: str,
+    user_state_dict: Optional[Dict[str, Any]] = None,
+    strict: bool = True,
+) -> Dict[str, Any]:
+    tracker = LatencyTracker(f"get_state_dict[{key}]")
+    try:
+        mapping = await client.get(f"{key}/{MAPPING_KEY}")
+    except KeyError as exc:
+        raise RuntimeError(
+            f"no state_dict was pushed under {key!r} (missing commit marker)"
+        ) from exc
+    tracker.step("mapping")
+
+    if user_state_dict is not None:
+        user_flat, user_mapping = _flatten(user_state_dict)
+        if strict:
+            stored_keys = {
+                str(k) for k in mapping
+            }
+            missing = set(user_flat.keys()) - stored_keys
+            if missing:
+                raise KeyError(
+                    f"user state_dict has entries not in the stored one: "
+                    f"{sorted(missing)[:5]}..."
+                )
+        fetches = {f"{key}/{k}": v for k, v in user_flat.items()}
+        results = await client.get_batch(fetches)
+        tracker.step("get_batch", _nbytes(user_flat))
+        flat = {k: results[f"{key}/{k}"] for k in user_flat}
+        out = _unflatten(flat, user_mapping)
+        tracker.e2e(_nbytes(user_flat))
+        return out
+
+    flat_keys = list(mapping.keys())
+    fetches = {f"{key}/{k}": None for k in flat_keys}
+    results = await client.get_batch(fetches)
+    tracker.step("get_batch")
+    flat = {k: results[f"{key}/{k}"] for k in flat_keys}
+    out = _unflatten(flat, mapping)
+    tracker.e2e(_nbytes(flat))
+    return out

@@ -159,7 +159,7 @@ class Request:
 
     @property
     def has_payload(self) -> bool:
-        return self.tensor_val is not None or self.is_object
+        return self.tensor_val is not None or self.objects is not None
 
     def nbytes(self) -> int:
         if self.tensor_val is not None:
