@@ -63,16 +63,18 @@ async def initialize(
     if num_storage_volumes is None:
         num_storage_volumes = 1
 
-    mesh = await asyncio.to_thread(
-        spawn_actors,
-        num_storage_volumes,
-        StorageVolume,
-        f"{store_name}-volume",
-        volume_id_seed=strategy.volume_id_seed,
-        device=storage_device,
-    )
-    controller = await asyncio.to_thread(
-        spawn_actor, Controller, f"{store_name}-controller", store_name
+    mesh, controller = await asyncio.gather(
+        asyncio.to_thread(
+            spawn_actors,
+            num_storage_volumes,
+            StorageVolume,
+            f"{store_name}-volume",
+            volume_id_seed=strategy.volume_id_seed,
+            device=storage_device,
+        ),
+        asyncio.to_thread(
+            spawn_actor, Controller, f"{store_name}-controller", store_name
+        ),
     )
     ids = await mesh.get_id.call()
     infos = [

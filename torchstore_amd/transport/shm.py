@@ -1,25 +1,28 @@
 """Same-host transport over POSIX shared memory.
 
 MI355X adaptation of the reference SHM transport (torchstore
-``transport/shared_memory.py``):
+``transport/shared_memory.py``) with one deliberate ownership inversion:
+**every segment is allocated by the (long-lived) storage volume**; clients
+only attach.  torch's filename-SHM manager makes the *allocating* process
+block on exit while peers hold attachments (~30 s), so short-lived client
+processes must never own segments.  A put therefore does a handshake RPC in
+which the volume allocates (or reuses, keyed by key) the segments and
+returns descriptors; the client attaches and copies in; the data RPC then
+just tells the volume to adopt the bytes.
 
-* segments are torch filename-shared CPU storages
-  (``UntypedStorage._new_using_filename_cpu``), attached on the peer by
-  ``_new_shared_filename_cpu`` — zero serialization of payload bytes;
-* GPU↔segment copies run on **dedicated per-device HIP copy streams** and
+Other properties:
+
+* GPU↔segment copies run on dedicated per-device HIP copy streams and
   synchronize only those streams (warm puts never stall unrelated streams —
   the reference's stream-isolation invariant, ``test_shared_memory.py:1034``);
 * segments touched by GPU copies are page-pinned via ``hipHostRegister``
   (``torch.cuda.cudart()`` maps to HIP on ROCm), fail-open with a
   once-per-error warning;
-* both sides keep caches: the client reuses put segments per key and attach
-  mappings per segment name; the volume reuses attach mappings, records
-  which stored tensors *are* segments (so warm gets of SHM-stored keys are
-  volume-side zero-copy), and reuses response segments per (key, region).
-
-Unlike the reference there is no handshake RPC: descriptors ride the data
-RPC and both sides' caches make reuse decisions locally — one round trip
-fewer per operation.
+* a CPU volume *adopts* the typed view of its own segment as storage, so a
+  warm get of an SHM-stored key returns the descriptor with **zero**
+  volume-side copies;
+* both sides cache attach mappings by segment name; the volume reuses
+  response segments per (key, region).
 """
 
 from __future__ import annotations
@@ -57,15 +60,18 @@ class ShmDescriptor:
     def seg_key(self) -> Tuple[bytes, bytes]:
         return (self.manager, self.name)
 
+    def with_layout(self, dtype: torch.dtype, shape) -> "ShmDescriptor":
+        return ShmDescriptor(self.manager, self.name, self.nbytes, dtype, tuple(shape))
+
 
 def _allocate_segment(nbytes: int) -> torch.Tensor:
     storage = torch.UntypedStorage._new_using_filename_cpu(nbytes)
     return torch.empty(0, dtype=torch.uint8).set_(storage)
 
 
-def _segment_handle(seg: torch.Tensor) -> Tuple[bytes, bytes, int]:
+def _segment_descriptor(seg: torch.Tensor) -> ShmDescriptor:
     manager, name, size = seg.untyped_storage()._share_filename_cpu_()
-    return manager, name, size
+    return ShmDescriptor(manager, name, size, torch.uint8, (size,))
 
 
 def _attach_segment(desc: ShmDescriptor) -> torch.Tensor:
@@ -81,9 +87,8 @@ def _typed_view(seg_u8: torch.Tensor, desc: ShmDescriptor) -> torch.Tensor:
         numel *= s
     if numel == 0:
         return torch.empty(desc.shape, dtype=desc.dtype)
-    return seg_u8[: numel * torch._utils._element_size(desc.dtype)].view(
-        desc.dtype
-    ).reshape(desc.shape)
+    esize = torch._utils._element_size(desc.dtype)
+    return seg_u8[: numel * esize].view(desc.dtype).reshape(desc.shape)
 
 
 def _try_pin(seg_u8: torch.Tensor, pinned: Dict[int, int]) -> None:
@@ -138,117 +143,175 @@ class _CopyStreams:
         return s
 
 
-def _copy_bytes(dst_u8: torch.Tensor, src: torch.Tensor, streams: _CopyStreams) -> Optional[torch.cuda.Stream]:
-    """Copy ``src`` (any device/dtype, contiguous) into a CPU byte span.
-
-    Returns the stream used (caller synchronizes it) or None for CPU→CPU.
-    """
+def _copy_into_u8(
+    dst_u8: torch.Tensor, src: torch.Tensor, streams: _CopyStreams
+) -> Optional[torch.cuda.Stream]:
+    """Copy ``src`` (any device, made contiguous) into a CPU byte span."""
     src_c = src.contiguous()
     src_u8 = byte_view(src_c)
     if src.device.type == "cuda":
         stream = streams.get(src.device.index)
-        # order after pending default-stream work that may produce src
         stream.wait_stream(torch.cuda.current_stream(src.device))
         with torch.cuda.stream(stream):
             dst_u8[: src_u8.numel()].copy_(src_u8, non_blocking=True)
-            # keep a possibly-temporary contiguous src alive for the copy
             src_c.record_stream(stream)
         return stream
     dst_u8[: src_u8.numel()].copy_(src_u8)
     return None
 
 
+def _copy_from_u8(
+    dest: torch.Tensor, typed_src: torch.Tensor, streams: _CopyStreams
+) -> Optional[torch.cuda.Stream]:
+    """Copy a typed CPU view into dest (any device, may be strided)."""
+    if dest.device.type == "cuda":
+        stream = streams.get(dest.device.index)
+        stream.wait_stream(torch.cuda.current_stream(dest.device))
+        with torch.cuda.stream(stream):
+            dest.copy_(typed_src, non_blocking=True)
+        return stream
+    dest.copy_(typed_src)
+    return None
+
+
 class ShmClientCache(TransportCache):
     def __init__(self):
-        self.put_segments: Dict[str, Tuple[ShmDescriptor, torch.Tensor]] = {}
         self.attached: Dict[Tuple[bytes, bytes], torch.Tensor] = {}
         self.streams = _CopyStreams()
         self.pinned: Dict[int, int] = {}
 
+    def attach(self, desc: ShmDescriptor) -> torch.Tensor:
+        seg = self.attached.get(desc.seg_key)
+        if seg is None:
+            seg = _attach_segment(desc)
+            self.attached[desc.seg_key] = seg
+        return seg
+
     def drop_key(self, key: str) -> None:
-        self.put_segments.pop(key, None)
+        return None  # attachments are segment-level
 
     def close(self) -> None:
         _unpin_all(self.pinned)
-        self.put_segments.clear()
         self.attached.clear()
 
 
 class ShmVolumeCache(TransportCache):
+    """Volume side owns every segment."""
+
     def __init__(self):
-        self.attached: Dict[Tuple[bytes, bytes], torch.Tensor] = {}
-        self.desc_by_storage: Dict[int, ShmDescriptor] = {}
+        # key -> (descriptor, segment) for put targets
+        self.put_segments: Dict[str, Tuple[ShmDescriptor, torch.Tensor]] = {}
+        # (key, region) -> (descriptor, segment) for get responses
         self.get_segments: Dict[Any, Tuple[ShmDescriptor, torch.Tensor]] = {}
+        # storage ptr -> descriptor (zero-copy warm gets of adopted entries)
+        self.desc_by_storage: Dict[int, ShmDescriptor] = {}
         self.streams = _CopyStreams()
         self.pinned: Dict[int, int] = {}
 
+    def obtain(
+        self, table: Dict, cache_key, nbytes: int
+    ) -> Tuple[ShmDescriptor, torch.Tensor]:
+        entry = table.get(cache_key)
+        if entry is None or entry[0].nbytes < nbytes:
+            seg = _allocate_segment(max(nbytes, 1))
+            desc = _segment_descriptor(seg)
+            table[cache_key] = (desc, seg)
+            self.desc_by_storage[seg.untyped_storage().data_ptr()] = desc
+            return desc, seg
+        return entry
+
     def drop_key(self, key: str) -> None:
+        entry = self.put_segments.pop(key, None)
+        if entry is not None:
+            self.desc_by_storage.pop(
+                entry[1].untyped_storage().data_ptr(), None
+            )
         for k in [k for k in self.get_segments if k[0] == key]:
-            del self.get_segments[k]
+            desc, seg = self.get_segments.pop(k)
+            self.desc_by_storage.pop(seg.untyped_storage().data_ptr(), None)
 
     def close(self) -> None:
         _unpin_all(self.pinned)
-        self.attached.clear()
-        self.desc_by_storage.clear()
+        self.put_segments.clear()
         self.get_segments.clear()
+        self.desc_by_storage.clear()
 
 
 class ShmTransportBuffer(TransportBuffer):
     transport_type = TransportType.SHARED_MEMORY
-    requires_handshake = False
+    requires_handshake = False  # put() below runs its own handshake phase
 
     def __init__(self):
         super().__init__()
-        # aligned with requests: ("obj", value) | ("shm", ShmDescriptor)
+        # put: aligned with requests, ("seg", ShmDescriptor) | ("obj", value)
+        # (descriptors are produced volume-side in the handshake)
         self.payload: Optional[List[Tuple[str, Any]]] = None
+        # put handshake: per-request nbytes to allocate (None for objects)
+        self.alloc_sizes: Optional[List[Optional[int]]] = None
+
+    # -- handshake (volume allocates put segments) ------------------------
+    def recv_handshake(self, requests: Sequence[Request], phase: str, volume):
+        if phase != "put":
+            return None
+        cache: ShmVolumeCache = self._volume_ctx.cache(ShmVolumeCache)
+        out: List[Optional[ShmDescriptor]] = []
+        for r, nbytes in zip(requests, self.alloc_sizes):
+            if nbytes is None:
+                out.append(None)
+                continue
+            desc, _seg = cache.obtain(cache.put_segments, r.key, nbytes)
+            out.append(desc)
+        return out
 
     # -- client put -------------------------------------------------------
-    async def client_stage_put(self, requests: Sequence[Request]) -> None:
-        cache: ShmClientCache = self._client_ctx.cache(ShmClientCache)
-        payload: List[Tuple[str, Any]] = []
-        streams_used = []
-        for r in requests:
-            if r.is_object:
-                payload.append(("obj", r.objects))
-                continue
-            t = r.tensor_val
-            nbytes = t.numel() * t.element_size()
-            entry = cache.put_segments.get(r.key)
-            if entry is None or entry[0].nbytes < nbytes:
-                seg = _allocate_segment(max(nbytes, 1))
-                manager, name, size = _segment_handle(seg)
-                desc = ShmDescriptor(manager, name, size, t.dtype, tuple(t.shape))
-                cache.put_segments[r.key] = (desc, seg)
-            else:
-                desc, seg = entry
-                desc = ShmDescriptor(
-                    desc.manager, desc.name, desc.nbytes, t.dtype, tuple(t.shape)
+    async def put(self, requests: Sequence[Request]) -> None:
+        volume = self._volume_ref.volume
+        metas = [r.meta_only() for r in requests]
+        self.alloc_sizes = [
+            None if r.is_object else r.nbytes() for r in requests
+        ]
+        try:
+            reply = await volume.handshake.call_one(self, metas, "put")
+            cache: ShmClientCache = self._client_ctx.cache(ShmClientCache)
+            payload: List[Tuple[str, Any]] = []
+            streams_used = []
+            for r, desc in zip(requests, reply):
+                if r.is_object or desc is None:
+                    payload.append(("obj", r.objects))
+                    continue
+                t = r.tensor_val
+                seg = cache.attach(desc)
+                if t.device.type == "cuda":
+                    _try_pin(seg, cache.pinned)
+                stream = _copy_into_u8(seg, t, cache.streams)
+                if stream is not None:
+                    streams_used.append(stream)
+                payload.append(
+                    ("seg", desc.with_layout(t.dtype, t.shape))
                 )
-                cache.put_segments[r.key] = (desc, seg)
-            if t.device.type == "cuda":
-                _try_pin(seg, cache.pinned)
-            stream = _copy_bytes(seg, t, cache.streams)
-            if stream is not None:
-                streams_used.append(stream)
-            payload.append(("shm", desc))
-        for s in set(streams_used):
-            s.synchronize()
-        self.payload = payload
+            for s in set(streams_used):
+                s.synchronize()
+            self.payload = payload
+            await volume.put.call_one(self, [r.meta_only() for r in requests])
+        finally:
+            await self.drop()
 
     # -- volume put -------------------------------------------------------
     async def volume_receive(self, requests, existing, device):
         cache: ShmVolumeCache = self._volume_ctx.cache(ShmVolumeCache)
         out: List[Any] = []
-        for (kind, value), prior in zip(self.payload, existing):
+        streams_used = []
+        for r, (kind, value), prior in zip(requests, self.payload, existing):
             if kind == "obj":
                 out.append(value)
                 continue
             desc: ShmDescriptor = value
-            seg = cache.attached.get(desc.seg_key)
-            if seg is None:
+            cached = cache.put_segments.get(r.key)
+            if cached is not None and cached[0].seg_key == desc.seg_key:
+                seg = cached[1]
+            else:
+                # shouldn't happen (handshake allocated it) — re-attach safely
                 seg = _attach_segment(desc)
-                cache.attached[desc.seg_key] = seg
-                cache.desc_by_storage[seg.untyped_storage().data_ptr()] = desc
             typed = _typed_view(seg, desc)
             if device.type == "cuda":
                 _try_pin(seg, cache.pinned)
@@ -258,15 +321,17 @@ class ShmTransportBuffer(TransportBuffer):
                     and prior.dtype == typed.dtype
                     and prior.device == device
                 ):
-                    prior.copy_(typed, non_blocking=True)
-                    torch.cuda.synchronize(device)
+                    stream = _copy_from_u8(prior, typed, cache.streams)
+                    if stream is not None:
+                        streams_used.append(stream)
                     out.append(prior)
                 else:
-                    gpu_t = typed.to(device)
-                    out.append(gpu_t)
+                    out.append(typed.to(device))
             else:
-                # CPU store adopts the segment — zero-copy warm path
+                # CPU store adopts the volume-owned segment — zero-copy
                 out.append(typed)
+        for s in set(streams_used):
+            s.synchronize()
         return out
 
     # -- volume get -------------------------------------------------------
@@ -279,39 +344,28 @@ class ShmTransportBuffer(TransportBuffer):
                 reply.append(("obj", v))
                 continue
             nbytes = v.numel() * v.element_size()
-            # zero-copy: the stored tensor IS a full attached segment
-            if v.device.type == "cpu" and v.is_contiguous() and v.storage_offset() == 0:
+            # zero-copy: the stored tensor IS (a prefix of) an owned segment
+            if (
+                v.device.type == "cpu"
+                and v.is_contiguous()
+                and v.storage_offset() == 0
+            ):
                 known = cache.desc_by_storage.get(v.untyped_storage().data_ptr())
                 if known is not None and known.nbytes >= nbytes:
-                    reply.append(
-                        ("shm", ShmDescriptor(
-                            known.manager, known.name, known.nbytes,
-                            v.dtype, tuple(v.shape),
-                        ))
-                    )
+                    reply.append(("seg", known.with_layout(v.dtype, v.shape)))
                     continue
             region = (
                 r.key,
                 r.tensor_slice.offsets if r.tensor_slice else None,
                 tuple(v.shape),
             )
-            entry = cache.get_segments.get(region)
-            if entry is None or entry[0].nbytes < nbytes:
-                seg = _allocate_segment(max(nbytes, 1))
-                manager, name, size = _segment_handle(seg)
-                desc = ShmDescriptor(manager, name, size, v.dtype, tuple(v.shape))
-                cache.get_segments[region] = (desc, seg)
-            else:
-                desc, seg = entry
-                desc = ShmDescriptor(
-                    desc.manager, desc.name, desc.nbytes, v.dtype, tuple(v.shape)
-                )
+            desc, seg = cache.obtain(cache.get_segments, region, nbytes)
             if v.device.type == "cuda":
                 _try_pin(seg, cache.pinned)
-            stream = _copy_bytes(seg, v, cache.streams)
+            stream = _copy_into_u8(seg, v, cache.streams)
             if stream is not None:
                 streams_used.append(stream)
-            reply.append(("shm", desc))
+            reply.append(("seg", desc.with_layout(v.dtype, v.shape)))
         for s in set(streams_used):
             s.synchronize()
         return reply
@@ -326,10 +380,7 @@ class ShmTransportBuffer(TransportBuffer):
                 out.append(value)
                 continue
             desc: ShmDescriptor = value
-            seg = cache.attached.get(desc.seg_key)
-            if seg is None:
-                seg = _attach_segment(desc)
-                cache.attached[desc.seg_key] = seg
+            seg = cache.attach(desc)
             typed = _typed_view(seg, desc)
             dest = r.tensor_val
             if dest is None:
@@ -337,13 +388,9 @@ class ShmTransportBuffer(TransportBuffer):
                 continue
             if dest.device.type == "cuda":
                 _try_pin(seg, cache.pinned)
-                stream = cache.streams.get(dest.device.index)
-                stream.wait_stream(torch.cuda.current_stream(dest.device))
-                with torch.cuda.stream(stream):
-                    dest.copy_(typed, non_blocking=True)
+            stream = _copy_from_u8(dest, typed, cache.streams)
+            if stream is not None:
                 streams_used.append(stream)
-            else:
-                dest.copy_(typed)
             out.append(dest)
         for s in set(streams_used):
             s.synchronize()
