@@ -1,0 +1,450 @@
+// _hipstore — native core of torchstore_amd for MI355X (gfx950, CDNA4).
+//
+// Provides (python surface wrapped by torchstore_amd/ops/gpu.py):
+//   * HIP IPC:  ipc_export / ipc_open / ipc_close — hipIpcMemHandle_t at
+//     caching-allocator block granularity (base via hipMemGetAddressRange,
+//     descriptor carries the byte offset), replacing the reference's
+//     ibverbs RDMA registration (torchstore transport/monarch_rdma.py).
+//   * copy_batch — batched one-sided bulk copies (hipMemcpyPeerAsync /
+//     DtoD) striped round-robin over a per-device pool of dedicated HIP
+//     streams so concurrent transfers to different peers aggregate xGMI
+//     links (7 x ~153 GB/s per GPU); synchronizes the streams it used.
+//   * copy_slices — K1/K2: one kernel launch copying N strided slices
+//     (gather: strided->contiguous, scatter: contiguous->strided, or
+//     strided->strided), replacing per-slice torch copies in reshard
+//     pack/unpack (reference hot loops: storage_volume.py:220-277,
+//     utils.py:199-212, direct_weight_sync.py:350).
+//   * cast_copy — K3: fused dtype cast + pack (f32<->bf16/f16), one HBM
+//     read + one write, vectorized 16B/lane (reference:
+//     state_dict_utils.py:177-189 casts every floating param per sync).
+//   * host_register / host_unregister — pin SHM pages for DMA.
+//
+// Deliberately torch-ABI-free: tensors cross as raw device pointers +
+// geometry, so the module builds with plain hipcc + pybind11 and never
+// drifts against libtorch.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <algorithm>
+#include <cstdint>
+#include <cstring>
+#include <mutex>
+#include <stdexcept>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+namespace py = pybind11;
+
+#define HIP_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      throw std::runtime_error(std::string(#expr) + " failed: " +              \
+                               hipGetErrorString(_e));                         \
+    }                                                                          \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// stream pools: dedicated streams per device for bulk copies
+// ---------------------------------------------------------------------------
+
+static constexpr int kStreamsPerDevice = 8;
+
+struct DevicePool {
+  std::vector<hipStream_t> streams;
+  // pinned host staging for slice descriptors + device copy of them
+  void* h_desc = nullptr;
+  void* d_desc = nullptr;
+  size_t desc_cap = 0;
+  // guards h_desc reuse: recorded after each staging H2D enqueue
+  hipEvent_t desc_evt = nullptr;
+};
+
+static std::mutex g_mutex;
+static std::unordered_map<int, DevicePool> g_pools;
+
+static DevicePool& pool_for(int device) {
+  std::lock_guard<std::mutex> lock(g_mutex);
+  auto it = g_pools.find(device);
+  if (it != g_pools.end()) return it->second;
+  HIP_CHECK(hipSetDevice(device));
+  DevicePool p;
+  p.streams.resize(kStreamsPerDevice);
+  for (int i = 0; i < kStreamsPerDevice; ++i) {
+    HIP_CHECK(hipStreamCreateWithFlags(&p.streams[i], hipStreamNonBlocking));
+  }
+  return g_pools.emplace(device, std::move(p)).first->second;
+}
+
+static void ensure_desc_capacity(DevicePool& p, int device, size_t bytes) {
+  if (p.desc_cap >= bytes) return;
+  size_t cap = bytes * 2 + 4096;
+  HIP_CHECK(hipSetDevice(device));
+  if (p.h_desc) HIP_CHECK(hipHostFree(p.h_desc));
+  if (p.d_desc) HIP_CHECK(hipFree(p.d_desc));
+  HIP_CHECK(hipHostMalloc(&p.h_desc, cap, hipHostMallocDefault));
+  HIP_CHECK(hipMalloc(&p.d_desc, cap));
+  p.desc_cap = cap;
+}
+
+// ---------------------------------------------------------------------------
+// HIP IPC
+// ---------------------------------------------------------------------------
+
+static py::tuple ipc_export(uintptr_t ptr, int device) {
+  HIP_CHECK(hipSetDevice(device));
+  void* base = nullptr;
+  size_t size = 0;
+  HIP_CHECK(hipMemGetAddressRange(&base, &size, reinterpret_cast<void*>(ptr)));
+  hipIpcMemHandle_t handle;
+  HIP_CHECK(hipIpcGetMemHandle(&handle, base));
+  uintptr_t offset = ptr - reinterpret_cast<uintptr_t>(base);
+  return py::make_tuple(
+      py::bytes(reinterpret_cast<const char*>(&handle), sizeof(handle)),
+      static_cast<uint64_t>(offset));
+}
+
+static uintptr_t ipc_open(py::bytes handle_bytes, int local_device,
+                          int src_device) {
+  std::string raw = handle_bytes;
+  if (raw.size() != sizeof(hipIpcMemHandle_t)) {
+    throw std::runtime_error("bad ipc handle size");
+  }
+  hipIpcMemHandle_t handle;
+  std::memcpy(&handle, raw.data(), sizeof(handle));
+  HIP_CHECK(hipSetDevice(local_device));
+  void* ptr = nullptr;
+  HIP_CHECK(hipIpcOpenMemHandle(&ptr, handle, hipIpcMemLazyEnablePeerAccess));
+  (void)src_device;
+  return reinterpret_cast<uintptr_t>(ptr);
+}
+
+static void ipc_close(uintptr_t base, int local_device) {
+  HIP_CHECK(hipSetDevice(local_device));
+  HIP_CHECK(hipIpcCloseMemHandle(reinterpret_cast<void*>(base)));
+}
+
+// ---------------------------------------------------------------------------
+// batched bulk copies over xGMI / HBM
+// ---------------------------------------------------------------------------
+
+// copies: (dst_ptr, dst_dev, src_ptr, src_dev, nbytes)
+static void copy_batch(
+    const std::vector<std::tuple<uintptr_t, int, uintptr_t, int, uint64_t>>&
+        copies) {
+  if (copies.empty()) return;
+  std::vector<hipStream_t> used;
+  int i = 0;
+  for (const auto& c : copies) {
+    uintptr_t dst = std::get<0>(c);
+    int dst_dev = std::get<1>(c);
+    uintptr_t src = std::get<2>(c);
+    int src_dev = std::get<3>(c);
+    uint64_t n = std::get<4>(c);
+    DevicePool& p = pool_for(dst_dev);
+    hipStream_t s = p.streams[i++ % kStreamsPerDevice];
+    HIP_CHECK(hipSetDevice(dst_dev));
+    if (dst_dev == src_dev) {
+      HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(dst),
+                               reinterpret_cast<void*>(src), n,
+                               hipMemcpyDeviceToDevice, s));
+    } else {
+      HIP_CHECK(hipMemcpyPeerAsync(reinterpret_cast<void*>(dst), dst_dev,
+                                   reinterpret_cast<void*>(src), src_dev, n,
+                                   s));
+    }
+    used.push_back(s);
+  }
+  for (hipStream_t s : used) HIP_CHECK(hipStreamSynchronize(s));
+}
+
+// ---------------------------------------------------------------------------
+// K1/K2 — batched strided slice copy
+// ---------------------------------------------------------------------------
+//
+// Each slice: rows of `row_bytes` contiguous bytes; row r's source/dest
+// offsets come from the outer-dims multi-index against byte strides.
+// Work unit = one TILE_BYTES span of one row; blocks grid-stride over the
+// global unit list and binary-search their slice in a prefix array.
+
+static constexpr int kMaxDims = 7;       // outer dims (innermost is the row)
+static constexpr uint32_t kTileBytes = 16384;  // 256 threads * 16B * 4 iters
+
+struct SliceDesc {
+  uintptr_t src;
+  uintptr_t dst;
+  uint64_t rows;            // product of outer dims
+  uint64_t units_prefix;    // exclusive prefix sum of units
+  uint32_t row_bytes;
+  uint32_t tiles_per_row;
+  uint32_t ndim;            // number of outer dims
+  uint32_t _pad;
+  uint64_t shape[kMaxDims];       // outer dims, innermost-last
+  int64_t src_stride[kMaxDims];   // byte strides of outer dims
+  int64_t dst_stride[kMaxDims];
+};
+
+__device__ __forceinline__ void row_offsets(const SliceDesc& d, uint64_t row,
+                                            int64_t& soff, int64_t& doff) {
+  soff = 0;
+  doff = 0;
+  uint64_t rem = row;
+  for (int i = (int)d.ndim - 1; i >= 0; --i) {
+    uint64_t idx = rem % d.shape[i];
+    rem /= d.shape[i];
+    soff += (int64_t)idx * d.src_stride[i];
+    doff += (int64_t)idx * d.dst_stride[i];
+  }
+}
+
+__global__ void __launch_bounds__(256)
+copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
+                   uint64_t total_units) {
+  for (uint64_t unit = blockIdx.x; unit < total_units; unit += gridDim.x) {
+    // binary search: greatest s with units_prefix <= unit
+    uint32_t lo = 0, hi = nslices - 1;
+    while (lo < hi) {
+      uint32_t mid = (lo + hi + 1) >> 1;
+      if (descs[mid].units_prefix <= unit) lo = mid; else hi = mid - 1;
+    }
+    const SliceDesc d = descs[lo];
+    uint64_t local = unit - d.units_prefix;
+    uint64_t row = local / d.tiles_per_row;
+    uint32_t tile = (uint32_t)(local % d.tiles_per_row);
+    int64_t soff, doff;
+    row_offsets(d, row, soff, doff);
+    uint32_t start = tile * kTileBytes;
+    uint32_t len = min(kTileBytes, d.row_bytes - start);
+    const char* src = reinterpret_cast<const char*>(d.src) + soff + start;
+    char* dst = reinterpret_cast<char*>(d.dst) + doff + start;
+    uintptr_t sa = reinterpret_cast<uintptr_t>(src);
+    uintptr_t da = reinterpret_cast<uintptr_t>(dst);
+    if (((sa | da | len) & 15u) == 0) {
+      // 16B vector path: coalesced dwordx4, 4 KiB per block iteration
+      const uint4* s4 = reinterpret_cast<const uint4*>(src);
+      uint4* d4 = reinterpret_cast<uint4*>(dst);
+      uint32_t n4 = len >> 4;
+      for (uint32_t i = threadIdx.x; i < n4; i += blockDim.x) d4[i] = s4[i];
+    } else if (((sa | da | len) & 3u) == 0) {
+      const uint32_t* s1 = reinterpret_cast<const uint32_t*>(src);
+      uint32_t* d1 = reinterpret_cast<uint32_t*>(dst);
+      uint32_t n1 = len >> 2;
+      for (uint32_t i = threadIdx.x; i < n1; i += blockDim.x) d1[i] = s1[i];
+    } else if (((sa | da | len) & 1u) == 0) {
+      const uint16_t* s1 = reinterpret_cast<const uint16_t*>(src);
+      uint16_t* d1 = reinterpret_cast<uint16_t*>(dst);
+      uint32_t n1 = len >> 1;
+      for (uint32_t i = threadIdx.x; i < n1; i += blockDim.x) d1[i] = s1[i];
+    } else {
+      for (uint32_t i = threadIdx.x; i < len; i += blockDim.x) dst[i] = src[i];
+    }
+  }
+}
+
+// python passes per slice:
+//   (src_ptr, dst_ptr, row_bytes, [outer shape], [src strides B], [dst strides B])
+using PySlice = std::tuple<uintptr_t, uintptr_t, uint64_t,
+                           std::vector<uint64_t>, std::vector<int64_t>,
+                           std::vector<int64_t>>;
+
+static void copy_slices(const std::vector<PySlice>& slices, int device,
+                        uintptr_t stream_handle, bool blocking) {
+  if (slices.empty()) return;
+  size_t n = slices.size();
+  std::vector<SliceDesc> descs(n);
+  uint64_t units = 0;
+  for (size_t i = 0; i < n; ++i) {
+    const auto& s = slices[i];
+    SliceDesc& d = descs[i];
+    d.src = std::get<0>(s);
+    d.dst = std::get<1>(s);
+    uint64_t row_bytes = std::get<2>(s);
+    const auto& shape = std::get<3>(s);
+    const auto& sst = std::get<4>(s);
+    const auto& dst = std::get<5>(s);
+    if (shape.size() > kMaxDims) throw std::runtime_error("too many dims");
+    if (row_bytes > UINT32_MAX) throw std::runtime_error("row too large");
+    d.ndim = (uint32_t)shape.size();
+    d.rows = 1;
+    for (size_t k = 0; k < shape.size(); ++k) {
+      d.shape[k] = shape[k];
+      d.src_stride[k] = sst[k];
+      d.dst_stride[k] = dst[k];
+      d.rows *= shape[k];
+    }
+    d.row_bytes = (uint32_t)row_bytes;
+    d.tiles_per_row = (uint32_t)((row_bytes + kTileBytes - 1) / kTileBytes);
+    if (d.tiles_per_row == 0) d.tiles_per_row = 1;
+    d.units_prefix = units;
+    units += d.rows * d.tiles_per_row;
+  }
+  if (units == 0) return;
+
+  DevicePool& p = pool_for(device);
+  HIP_CHECK(hipSetDevice(device));
+  size_t bytes = n * sizeof(SliceDesc);
+  hipStream_t stream = reinterpret_cast<hipStream_t>(stream_handle);
+  uint32_t grid = 0;
+  {
+    std::lock_guard<std::mutex> lock(g_mutex);
+    ensure_desc_capacity(p, device, bytes);
+    if (p.desc_evt == nullptr) {
+      HIP_CHECK(hipEventCreateWithFlags(&p.desc_evt, hipEventDisableTiming));
+    } else {
+      // previous call's staging H2D must be done before h_desc is reused
+      HIP_CHECK(hipEventSynchronize(p.desc_evt));
+    }
+    std::memcpy(p.h_desc, descs.data(), bytes);
+    HIP_CHECK(hipMemcpyAsync(p.d_desc, p.h_desc, bytes, hipMemcpyHostToDevice,
+                             stream));
+    HIP_CHECK(hipEventRecord(p.desc_evt, stream));
+    // memory-bound: cap grid at 2048 blocks, grid-stride the rest (guide G11)
+    grid = (uint32_t)std::min<uint64_t>(units, 2048);
+    hipLaunchKernelGGL(copy_slices_kernel, dim3(grid), dim3(256), 0, stream,
+                       reinterpret_cast<const SliceDesc*>(p.d_desc),
+                       (uint32_t)n, units);
+    HIP_CHECK(hipGetLastError());
+  }
+  if (blocking) HIP_CHECK(hipStreamSynchronize(stream));
+}
+
+// ---------------------------------------------------------------------------
+// K3 — fused cast + pack
+// ---------------------------------------------------------------------------
+
+enum class DType : int32_t { F32 = 0, F16 = 1, BF16 = 2 };
+
+template <typename SrcT, typename DstT>
+__device__ __forceinline__ DstT convert(SrcT v);
+
+template <> __device__ __forceinline__ float convert<float, float>(float v) { return v; }
+template <> __device__ __forceinline__ __hip_bfloat16 convert<float, __hip_bfloat16>(float v) {
+  return __float2bfloat16(v);
+}
+template <> __device__ __forceinline__ float convert<__hip_bfloat16, float>(__hip_bfloat16 v) {
+  return __bfloat162float(v);
+}
+template <> __device__ __forceinline__ __half convert<float, __half>(float v) {
+  return __float2half(v);
+}
+template <> __device__ __forceinline__ float convert<__half, float>(__half v) {
+  return __half2float(v);
+}
+template <> __device__ __forceinline__ __half convert<__hip_bfloat16, __half>(__hip_bfloat16 v) {
+  return __float2half(__bfloat162float(v));
+}
+template <> __device__ __forceinline__ __hip_bfloat16 convert<__half, __hip_bfloat16>(__half v) {
+  return __float2bfloat16(__half2float(v));
+}
+
+// 8 elements per thread per iteration; 16B loads when SrcT is 2 bytes,
+// 32B (2x float4) when 4 bytes — always >= the 16B/lane coalescing sweet
+// spot on the wider side (guide G13).
+template <typename SrcT, typename DstT>
+__global__ void __launch_bounds__(256)
+cast_copy_kernel(const SrcT* __restrict__ src, DstT* __restrict__ dst,
+                 uint64_t numel) {
+  constexpr uint32_t V = 8;
+  uint64_t base = (uint64_t)(blockIdx.x * blockDim.x + threadIdx.x) * V;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x * V;
+  struct alignas(16) SrcVec { SrcT v[V]; };
+  struct alignas(16) DstVec { DstT v[V]; };
+  for (uint64_t i = base; i + V <= numel; i += stride) {
+    SrcVec s = *reinterpret_cast<const SrcVec*>(src + i);
+    DstVec d;
+#pragma unroll
+    for (uint32_t k = 0; k < V; ++k) d.v[k] = convert<SrcT, DstT>(s.v[k]);
+    *reinterpret_cast<DstVec*>(dst + i) = d;
+  }
+  // tail
+  uint64_t tail_start = (numel / V) * V;
+  uint64_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid < numel - tail_start) {
+    uint64_t i = tail_start + tid;
+    dst[i] = convert<SrcT, DstT>(src[i]);
+  }
+}
+
+template <typename SrcT, typename DstT>
+static void launch_cast(uintptr_t src, uintptr_t dst, uint64_t numel,
+                        hipStream_t stream) {
+  uint64_t work = (numel + 7) / 8;
+  uint32_t grid = (uint32_t)std::min<uint64_t>((work + 255) / 256, 2048);
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL((cast_copy_kernel<SrcT, DstT>), dim3(grid), dim3(256), 0,
+                     stream, reinterpret_cast<const SrcT*>(src),
+                     reinterpret_cast<DstT*>(dst), numel);
+  HIP_CHECK(hipGetLastError());
+}
+
+static void cast_copy(uintptr_t src, int src_dtype, uintptr_t dst,
+                      int dst_dtype, uint64_t numel, int device,
+                      uintptr_t stream_handle) {
+  HIP_CHECK(hipSetDevice(device));
+  hipStream_t stream = reinterpret_cast<hipStream_t>(stream_handle);
+  auto s = static_cast<DType>(src_dtype);
+  auto d = static_cast<DType>(dst_dtype);
+  if (s == DType::F32 && d == DType::BF16)
+    launch_cast<float, __hip_bfloat16>(src, dst, numel, stream);
+  else if (s == DType::BF16 && d == DType::F32)
+    launch_cast<__hip_bfloat16, float>(src, dst, numel, stream);
+  else if (s == DType::F32 && d == DType::F16)
+    launch_cast<float, __half>(src, dst, numel, stream);
+  else if (s == DType::F16 && d == DType::F32)
+    launch_cast<__half, float>(src, dst, numel, stream);
+  else if (s == DType::BF16 && d == DType::F16)
+    launch_cast<__hip_bfloat16, __half>(src, dst, numel, stream);
+  else if (s == DType::F16 && d == DType::BF16)
+    launch_cast<__half, __hip_bfloat16>(src, dst, numel, stream);
+  else
+    throw std::runtime_error("unsupported cast pair");
+}
+
+// ---------------------------------------------------------------------------
+// host pinning
+// ---------------------------------------------------------------------------
+
+static void host_register(uintptr_t ptr, uint64_t nbytes) {
+  HIP_CHECK(hipHostRegister(reinterpret_cast<void*>(ptr), nbytes,
+                            hipHostRegisterPortable));
+}
+
+static void host_unregister(uintptr_t ptr) {
+  HIP_CHECK(hipHostUnregister(reinterpret_cast<void*>(ptr)));
+}
+
+static int device_count() {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  return e == hipSuccess ? n : 0;
+}
+
+static void sync_device(int device) {
+  HIP_CHECK(hipSetDevice(device));
+  HIP_CHECK(hipDeviceSynchronize());
+}
+
+PYBIND11_MODULE(_hipstore, m) {
+  m.doc() = "torchstore_amd native core (HIP/CDNA4, gfx950)";
+  m.def("ipc_export", &ipc_export, py::arg("ptr"), py::arg("device"));
+  m.def("ipc_open", &ipc_open, py::arg("handle"), py::arg("local_device"),
+        py::arg("src_device"));
+  m.def("ipc_close", &ipc_close, py::arg("base"), py::arg("local_device"));
+  m.def("copy_batch", &copy_batch, py::arg("copies"),
+        py::call_guard<py::gil_scoped_release>());
+  m.def("copy_slices", &copy_slices, py::arg("slices"), py::arg("device"),
+        py::arg("stream"), py::arg("blocking") = true,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("cast_copy", &cast_copy, py::arg("src"), py::arg("src_dtype"),
+        py::arg("dst"), py::arg("dst_dtype"), py::arg("numel"),
+        py::arg("device"), py::arg("stream"));
+  m.def("host_register", &host_register);
+  m.def("host_unregister", &host_unregister);
+  m.def("device_count", &device_count);
+  m.def("sync_device", &sync_device);
+}

@@ -60,3 +60,102 @@ def ext():
     if e is None:
         raise RuntimeError(f"_hipstore extension unavailable: {_ext_err}")
     return e
+
+
+# ---------------------------------------------------------------------------
+# typed wrappers (tensor → raw pointer marshalling)
+# ---------------------------------------------------------------------------
+
+_DTYPE_CODES = {torch.float32: 0, torch.float16: 1, torch.bfloat16: 2}
+
+
+def _stream(device: torch.device) -> int:
+    return torch.cuda.current_stream(device).cuda_stream
+
+
+def cast_copy(src: torch.Tensor, out: torch.Tensor) -> None:
+    """K3: fused cast+pack on the caller's current stream (async)."""
+    if src.numel() != out.numel():
+        raise ValueError("cast_copy numel mismatch")
+    if not (src.is_contiguous() and out.is_contiguous()):
+        raise ValueError("cast_copy requires contiguous tensors")
+    sc = _DTYPE_CODES.get(src.dtype)
+    dc = _DTYPE_CODES.get(out.dtype)
+    if (
+        sc is None
+        or dc is None
+        or src.data_ptr() % 16
+        or out.data_ptr() % 16
+    ):
+        # unusual dtype pair / misaligned view: defer to torch
+        # (K3 covers the hot float paths on allocator-aligned tensors)
+        out.copy_(src)
+        return
+    ext().cast_copy(
+        src.data_ptr(), sc, out.data_ptr(), dc,
+        src.numel(), src.device.index, _stream(src.device),
+    )
+
+
+def _slice_desc(src: torch.Tensor, dst: torch.Tensor):
+    """Build a copy_slices descriptor; None when the pair doesn't qualify."""
+    if src.shape != dst.shape or src.dtype != dst.dtype:
+        return None
+    if src.numel() == 0:
+        return ()
+    es = src.element_size()
+    if src.dim() == 0:
+        return (src.data_ptr(), dst.data_ptr(), es, [], [], [])
+    if src.stride(-1) != 1 or dst.stride(-1) != 1:
+        return None
+    row_bytes = src.shape[-1] * es
+    outer = list(src.shape[:-1])
+    sst = [s * es for s in src.stride()[:-1]]
+    dstst = [s * es for s in dst.stride()[:-1]]
+    return (src.data_ptr(), dst.data_ptr(), row_bytes, outer, sst, dstst)
+
+
+def copy_pairs(
+    pairs, device: torch.device, blocking: bool = True
+) -> None:
+    """K1/K2: batched strided copies ``[(src_view, dst_view), ...]``.
+
+    All tensors must be on ``device``.  Pairs whose layout the kernel can't
+    express (non-unit innermost stride) fall back to ``copy_``.
+    """
+    descs = []
+    for src, dst in pairs:
+        d = _slice_desc(src, dst)
+        if d is None:
+            dst.copy_(src)
+        elif d != ():
+            descs.append(d)
+    if descs:
+        ext().copy_slices(descs, device.index, _stream(device), blocking)
+
+
+def pack_region(src_view: torch.Tensor) -> torch.Tensor:
+    """K1: strided region → freshly-allocated contiguous tensor."""
+    if src_view.is_contiguous():
+        return src_view
+    out = torch.empty(
+        src_view.shape, dtype=src_view.dtype, device=src_view.device
+    )
+    copy_pairs([(src_view, out)], src_view.device, blocking=False)
+    return out
+
+
+def ipc_export(ptr: int, device_index: int):
+    return ext().ipc_export(ptr, device_index)
+
+
+def ipc_open(handle: bytes, local_device: int, src_device: int) -> int:
+    return ext().ipc_open(handle, local_device, src_device)
+
+
+def ipc_close(base: int, local_device: int) -> None:
+    ext().ipc_close(base, local_device)
+
+
+def copy_batch(copies) -> None:
+    ext().copy_batch(copies)

@@ -167,6 +167,7 @@ class HipIpcTransportBuffer(TransportBuffer):
     # ------------------------------------------------------------- get --
     async def client_stage_get(self, requests: Sequence[Request]) -> None:
         payload: List[Tuple[str, Any]] = []
+        synced: set = set()
         for i, r in enumerate(requests):
             if r.is_object:
                 payload.append(("fetch_obj", None))
@@ -184,6 +185,11 @@ class HipIpcTransportBuffer(TransportBuffer):
                     dest.shape, dtype=dest.dtype, device=dest.device
                 )
                 self._scratch[i] = target
+            if target.device.index not in synced:
+                # pending client kernels touching dest must finish before the
+                # volume's one-sided writes land in it
+                torch.cuda.current_stream(target.device).synchronize()
+                synced.add(target.device.index)
             self._hold.append(target)
             payload.append(("ipc", export_tensor(target)))
         self.payload = payload
@@ -202,7 +208,9 @@ class HipIpcTransportBuffer(TransportBuffer):
                 continue
             desc: IpcDescriptor = value
             device = v.device
-            vc = v.contiguous()  # K1 slice gather for strided stored views
+            from torchstore_amd.ops import gpu as gpu_ops
+
+            vc = gpu_ops.pack_region(v)  # K1 slice gather for strided views
             if vc.numel() * vc.element_size() != desc.nbytes:
                 raise RuntimeError(
                     f"get size mismatch for {r.key}: stored {vc.shape} vs "
@@ -215,6 +223,10 @@ class HipIpcTransportBuffer(TransportBuffer):
             )
             reply.append(("done", None))
             self._hold.append(vc)
+        if copies and device is not None:
+            # K1 pack kernels ran on the current stream; the pool streams
+            # used by copy_batch must observe their writes
+            torch.cuda.current_stream(device).synchronize()
         _run_copies(copies)
         return reply
 
