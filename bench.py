@@ -1,0 +1,180 @@
+"""Flagship benchmark: Llama-3-8B bf16 state_dict sync through the store.
+
+Measures the BASELINE.json headline on MI355X: each rank holds the
+FSDP-style Shard(0) layout of a random-init Llama-3-8B (~16.06 GB bf16),
+one **step** = push the full state_dict into GPU-resident storage volumes
+(put_state_dict) + pull it back in the TP-style layout (get_state_dict with
+resharding).  value = whole-job aggregate GB/s moved (put+get bytes / step
+time, max over ranks).  The reference (meta-pytorch/torchstore) publishes
+no numbers — this is the self-measured baseline.
+
+Single GPU:      python bench.py --steps 5 --warmup 2
+N GPUs (driver): python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                     --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=5)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--layers", type=int, default=None,
+                   help="override layer count (debug only; invalidates the metric)")
+    p.add_argument("--mode", choices=["state_dict", "direct"],
+                   default="state_dict")
+    return p.parse_args()
+
+
+def setup_dist(args):
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+        torch.distributed.init_process_group(
+            "nccl", rank=rank, world_size=world
+        )
+        torch.cuda.set_device(local_rank)
+    else:
+        if torch.cuda.is_available():
+            torch.cuda.set_device(0)
+    return rank, world, local_rank
+
+
+async def run_bench(args, rank, world, local_rank):
+    import torchstore_amd as ts
+    from torchstore_amd.models import llama
+    from torchstore_amd.strategy import LocalRankStrategy
+
+    device = f"cuda:{local_rank}"
+    mesh = None
+    if world > 1:
+        from torch.distributed.device_mesh import init_device_mesh
+
+        mesh = init_device_mesh("cuda", (world,))
+
+    layers = args.layers or llama.LAYERS
+    src_sd = llama.make_sharded_state_dict(
+        mesh, llama.fsdp_placement, device=device, layers=layers, seed=1234 + rank
+    )
+    dst_sd = llama.make_sharded_state_dict(
+        mesh, llama.tp_placement, device=device, layers=layers, zero=True
+    )
+    shapes = llama.llama3_8b_shapes(layers)
+    payload_bytes = llama.total_bytes(shapes, torch.bfloat16)
+
+    # store bring-up: rank 0 spawns volumes (one per GPU) + controller and
+    # publishes the controller handle through a TCPStore
+    if world > 1:
+        from torch.distributed import TCPStore
+
+        store = TCPStore(
+            os.environ["MASTER_ADDR"], int(os.environ["MASTER_PORT"]) + 1,
+            world, is_master=rank == 0,
+        )
+        if rank == 0:
+            controller = await ts.initialize(
+                num_storage_volumes=world,
+                strategy=LocalRankStrategy(),
+                storage_device="auto",
+            )
+            import pickle
+
+            store.set("controller", pickle.dumps(controller))
+        else:
+            import pickle
+
+            controller = pickle.loads(store.get("controller"))
+            ts.attach(controller, LocalRankStrategy())
+    else:
+        await ts.initialize(
+            num_storage_volumes=1,
+            strategy=LocalRankStrategy(),
+            storage_device="auto",
+        )
+
+    def barrier():
+        if world > 1:
+            torch.distributed.barrier()
+        torch.cuda.synchronize()
+
+    async def one_step():
+        await ts.put_state_dict(src_sd, "bench")
+        if world > 1:
+            torch.distributed.barrier()  # all shards committed
+        await ts.get_state_dict("bench", dst_sd)
+
+    for _ in range(args.warmup):
+        await one_step()
+    barrier()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        await one_step()
+    barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = t.item()
+
+    ms_per_step = elapsed / args.steps * 1e3
+    moved = 2 * payload_bytes  # put + get per step, whole job
+    gbps = moved / (elapsed / args.steps) / 1e9
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "llama3_8b_state_dict_sync_GBps",
+            "value": round(gbps, 2),
+            "unit": "GB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic (random-init Llama-3-8B shapes)",
+            "config": {
+                "model": "Llama-3-8B",
+                "global_batch": None,
+                "seq_len": None,
+                "parallelism": f"store fsdp{world}->tp{world}",
+                "payload_gb": round(payload_bytes / 1e9, 2),
+                "mode": args.mode,
+            },
+        }))
+
+    if world > 1:
+        torch.distributed.barrier()
+    if rank == 0 or world == 1:
+        await ts.shutdown()
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+def main():
+    args = parse_args()
+    if not torch.cuda.is_available():
+        raise SystemExit("bench.py needs MI355X GPUs (torch.cuda unavailable)")
+    rank, world, local_rank = setup_dist(args)
+    asyncio.run(run_bench(args, rank, world, local_rank))
+
+
+if __name__ == "__main__":
+    main()

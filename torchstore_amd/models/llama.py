@@ -1,0 +1,114 @@
+"""Synthetic Llama-3-8B state_dict (random init, no network access needed).
+
+Used by the bench + real-model tests: exact parameter shapes of
+meta-llama/Meta-Llama-3-8B (~8.03B params, ~16.06 GB bf16), materialized
+shard-by-shard so each rank only allocates its own slice.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+# (name template, shape) — 32 transformer layers
+HIDDEN = 4096
+INTER = 14336
+KV = 1024
+VOCAB = 128256
+LAYERS = 32
+
+
+def llama3_8b_shapes(layers: int = LAYERS) -> Dict[str, Tuple[int, ...]]:
+    shapes: Dict[str, Tuple[int, ...]] = {
+        "model.embed_tokens.weight": (VOCAB, HIDDEN),
+    }
+    for i in range(layers):
+        p = f"model.layers.{i}."
+        shapes[p + "self_attn.q_proj.weight"] = (HIDDEN, HIDDEN)
+        shapes[p + "self_attn.k_proj.weight"] = (KV, HIDDEN)
+        shapes[p + "self_attn.v_proj.weight"] = (KV, HIDDEN)
+        shapes[p + "self_attn.o_proj.weight"] = (HIDDEN, HIDDEN)
+        shapes[p + "mlp.gate_proj.weight"] = (INTER, HIDDEN)
+        shapes[p + "mlp.up_proj.weight"] = (INTER, HIDDEN)
+        shapes[p + "mlp.down_proj.weight"] = (HIDDEN, INTER)
+        shapes[p + "input_layernorm.weight"] = (HIDDEN,)
+        shapes[p + "post_attention_layernorm.weight"] = (HIDDEN,)
+    shapes["model.norm.weight"] = (HIDDEN,)
+    shapes["lm_head.weight"] = (VOCAB, HIDDEN)
+    return shapes
+
+
+def total_bytes(
+    shapes: Dict[str, Tuple[int, ...]], dtype: torch.dtype = torch.bfloat16
+) -> int:
+    esize = torch._utils._element_size(dtype)
+    return sum(
+        esize * int(torch.tensor(s).prod()) if s else esize
+        for s in shapes.values()
+    )
+
+
+def fsdp_placement(shape: Tuple[int, ...], world: int) -> Optional[int]:
+    """FSDP-style: Shard(0) when divisible, else replicate (None)."""
+    if len(shape) >= 1 and shape[0] % world == 0:
+        return 0
+    return None
+
+
+def tp_placement(shape: Tuple[int, ...], world: int) -> Optional[int]:
+    """TP-style: prefer Shard(1) on 2-D weights, else Shard(0), else replicate."""
+    if len(shape) == 2 and shape[1] % world == 0:
+        return 1
+    if len(shape) >= 1 and shape[0] % world == 0:
+        return 0
+    return None
+
+
+def make_sharded_state_dict(
+    mesh,
+    shard_fn,
+    dtype: torch.dtype = torch.bfloat16,
+    device: str = "cuda",
+    layers: int = LAYERS,
+    zero: bool = False,
+    seed: Optional[int] = None,
+) -> Dict[str, torch.Tensor]:
+    """Build {name: DTensor|tensor} where each rank materializes only its shard.
+
+    ``mesh`` is a 1-D DeviceMesh (or None for single-process: plain tensors).
+    ``shard_fn(shape, world) -> dim|None`` picks the shard dim per param.
+    """
+    from torch.distributed.tensor import DTensor, Replicate, Shard
+    from torch.distributed.tensor._utils import (
+        compute_local_shape_and_global_offset,
+    )
+
+    shapes = llama3_8b_shapes(layers)
+    out: Dict[str, torch.Tensor] = {}
+    world = mesh.size() if mesh is not None else 1
+    gen = None
+    if seed is not None:
+        gen = torch.Generator(device=device)
+        gen.manual_seed(seed)
+    for name, shape in shapes.items():
+        if mesh is None or world == 1:
+            t = torch.empty(shape, dtype=dtype, device=device)
+            if not zero:
+                t.normal_(0, 0.02, generator=gen)
+            out[name] = t
+            continue
+        dim = shard_fn(shape, world)
+        placements = [Replicate() if dim is None else Shard(dim)]
+        local_shape, _off = compute_local_shape_and_global_offset(
+            shape, mesh, placements
+        )
+        local = torch.empty(local_shape, dtype=dtype, device=device)
+        if not zero:
+            local.normal_(0, 0.02, generator=gen)
+        out[name] = DTensor.from_local(
+            local, mesh, placements, run_check=False,
+            shape=torch.Size(shape),
+            stride=torch.empty(shape, device="meta").stride(),
+        )
+    return out

@@ -18,9 +18,8 @@ def cast_tensor(t: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     if t.device.type == "cuda":
         from torchstore_amd.ops import gpu
 
-        ext = gpu.ext()
         out = torch.empty(t.shape, dtype=dtype, device=t.device)
-        ext.cast_copy(t, out)
+        gpu.cast_copy(t.contiguous(), out)
         return out
     return t.to(dtype)
 
@@ -30,6 +29,6 @@ def cast_into(t: torch.Tensor, out: torch.Tensor) -> None:
     if t.device.type == "cuda" and out.device.type == "cuda":
         from torchstore_amd.ops import gpu
 
-        gpu.ext().cast_copy(t, out)
+        gpu.cast_copy(t.contiguous(), out)
         return
     out.copy_(t)

@@ -232,6 +232,8 @@ class HipIpcTransportBuffer(TransportBuffer):
 
     def client_complete_get(self, requests, reply) -> List[Any]:
         out: List[Any] = []
+        scatter_pairs = []
+        device = None
         for i, (r, (kind, value)) in enumerate(zip(requests, reply)):
             if kind == "inline":
                 if r.tensor_val is not None and isinstance(value, torch.Tensor):
@@ -242,8 +244,14 @@ class HipIpcTransportBuffer(TransportBuffer):
                 continue
             scratch = self._scratch.get(i)
             if scratch is not None:
-                r.tensor_val.copy_(scratch)  # K2 scatter on the strided dest
+                scatter_pairs.append((scratch, r.tensor_val))
+                device = r.tensor_val.device
             out.append(r.tensor_val)
+        if scatter_pairs:
+            # K2: one batched scatter kernel for every strided destination
+            from torchstore_amd.ops import gpu as gpu_ops
+
+            gpu_ops.copy_pairs(scatter_pairs, device, blocking=False)
         return out
 
     async def drop(self) -> None:
