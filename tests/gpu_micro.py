@@ -1,0 +1,93 @@
+"""GPU microbench of the native primitives (run manually via gpurun).
+
+Times: ipc_export, copy_batch D2D, copy_slices pack, cast_copy — the
+building blocks of every transport operation.
+"""
+
+import time
+
+import torch
+
+from torchstore_amd.ops import gpu
+
+
+def timeit(label, fn, n=20, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(n):
+        fn()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / n
+    print(f"{label:40s} {dt*1e3:9.3f} ms")
+    return dt
+
+
+def main():
+    torch.cuda.set_device(0)
+    e = gpu.ext()
+
+    # export cost (fresh allocations vs same block)
+    t = torch.randn(1 << 20, device="cuda")
+    timeit("ipc_export same tensor", lambda: e.ipc_export(t.data_ptr(), 0))
+
+    ts = [torch.randn(1 << 18, device="cuda") for _ in range(64)]
+
+    def export_many():
+        for x in ts:
+            e.ipc_export(x.data_ptr(), 0)
+
+    dt = timeit("ipc_export x64 tensors", export_many, n=5)
+    print(f"  per export: {dt/64*1e3:.3f} ms")
+
+    # D2D copy bandwidth via copy_batch (same device)
+    for size_mb in [1, 64, 1024]:
+        n = size_mb * (1 << 20)
+        a = torch.empty(n, dtype=torch.uint8, device="cuda")
+        b = torch.empty(n, dtype=torch.uint8, device="cuda")
+        dt = timeit(
+            f"copy_batch 1x{size_mb}MB d2d",
+            lambda: e.copy_batch([(b.data_ptr(), 0, a.data_ptr(), 0, n)]),
+            n=10,
+        )
+        print(f"  bw: {n/dt/1e9:.1f} GB/s")
+
+    # batched: 64 copies of 16MB
+    srcs = [torch.empty(16 << 20, dtype=torch.uint8, device="cuda") for _ in range(64)]
+    dsts = [torch.empty(16 << 20, dtype=torch.uint8, device="cuda") for _ in range(64)]
+    copies = [
+        (d.data_ptr(), 0, s.data_ptr(), 0, 16 << 20) for s, d in zip(srcs, dsts)
+    ]
+    dt = timeit("copy_batch 64x16MB d2d", lambda: e.copy_batch(copies), n=10)
+    print(f"  bw: {64*(16<<20)/dt/1e9:.1f} GB/s")
+
+    # K1 pack bandwidth
+    big = torch.randn(8192, 8192, device="cuda")
+    view = big[:, 2048:6144]
+    out = torch.empty(view.shape, dtype=big.dtype, device="cuda")
+    dt = timeit(
+        "copy_slices pack 8192x4096 f32",
+        lambda: gpu.copy_pairs([(view, out)], big.device, blocking=True),
+        n=10,
+    )
+    nb = view.numel() * 4 * 2
+    print(f"  bw (rd+wr): {nb/dt/1e9:.1f} GB/s")
+    dt = timeit(
+        "torch .contiguous same view",
+        lambda: view.contiguous(),
+        n=10,
+    )
+    print(f"  bw (rd+wr): {nb/dt/1e9:.1f} GB/s")
+
+    # K3 cast bandwidth
+    f = torch.randn(1 << 28, device="cuda")  # 1 GiB fp32
+    o = torch.empty(1 << 28, dtype=torch.bfloat16, device="cuda")
+    dt = timeit("cast_copy 1GiB f32->bf16", lambda: gpu.cast_copy(f, o), n=10)
+    print(f"  bw (rd+wr): {(f.numel()*6)/dt/1e9:.1f} GB/s")
+    dt = timeit("torch .to(bf16) same", lambda: f.to(torch.bfloat16), n=10)
+    print(f"  bw (rd+wr): {(f.numel()*6)/dt/1e9:.1f} GB/s")
+
+
+if __name__ == "__main__":
+    main()

@@ -75,8 +75,18 @@ def resolve_transport_type(volume_ref) -> TransportType:
     return TransportType.RPC
 
 
+_logged_resolutions = set()
+
+
 def create_transport(volume_ref, ctx: Optional[TransportContext] = None) -> TransportBuffer:
     ttype = resolve_transport_type(volume_ref)
+    if (volume_ref.volume_id, ttype) not in _logged_resolutions:
+        _logged_resolutions.add((volume_ref.volume_id, ttype))
+        logger.info(
+            "transport to volume %s (%s, %s): %s",
+            volume_ref.volume_id, volume_ref.hostname, volume_ref.device,
+            ttype.value,
+        )
     reg = _registry()
     cls = reg.get(ttype)
     if cls is None:
