@@ -4,7 +4,11 @@ Times: ipc_export, copy_batch D2D, copy_slices pack, cast_copy — the
 building blocks of every transport operation.
 """
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -96,17 +96,43 @@ static void ensure_desc_capacity(DevicePool& p, int device, size_t bytes) {
 // HIP IPC
 // ---------------------------------------------------------------------------
 
+// Export handles are cached per allocator block (hipIpcGetMemHandle does a
+// dmabuf export ioctl — tens of ms — while hipMemGetAddressRange is cheap).
+// Same design as the reference's (data_ptr, nbytes)-keyed RdmaMemory cache
+// (torchstore torchcomms/cache.py:150-187).  The cache is only invalid if a
+// block is returned to the OS (torch.cuda.empty_cache) and the same base is
+// re-allocated — call ipc_export_cache_clear() around such events.
+static std::mutex g_export_mutex;
+static std::unordered_map<uintptr_t, std::string> g_export_cache;
+
 static py::tuple ipc_export(uintptr_t ptr, int device) {
   HIP_CHECK(hipSetDevice(device));
   void* base = nullptr;
   size_t size = 0;
   HIP_CHECK(hipMemGetAddressRange(&base, &size, reinterpret_cast<void*>(ptr)));
-  hipIpcMemHandle_t handle;
-  HIP_CHECK(hipIpcGetMemHandle(&handle, base));
-  uintptr_t offset = ptr - reinterpret_cast<uintptr_t>(base);
-  return py::make_tuple(
-      py::bytes(reinterpret_cast<const char*>(&handle), sizeof(handle)),
-      static_cast<uint64_t>(offset));
+  uintptr_t base_u = reinterpret_cast<uintptr_t>(base);
+  std::string handle_str;
+  {
+    std::lock_guard<std::mutex> lock(g_export_mutex);
+    auto it = g_export_cache.find(base_u);
+    if (it != g_export_cache.end()) {
+      handle_str = it->second;
+    }
+  }
+  if (handle_str.empty()) {
+    hipIpcMemHandle_t handle;
+    HIP_CHECK(hipIpcGetMemHandle(&handle, base));
+    handle_str.assign(reinterpret_cast<const char*>(&handle), sizeof(handle));
+    std::lock_guard<std::mutex> lock(g_export_mutex);
+    g_export_cache.emplace(base_u, handle_str);
+  }
+  return py::make_tuple(py::bytes(handle_str),
+                        static_cast<uint64_t>(ptr - base_u));
+}
+
+static void ipc_export_cache_clear() {
+  std::lock_guard<std::mutex> lock(g_export_mutex);
+  g_export_cache.clear();
 }
 
 static uintptr_t ipc_open(py::bytes handle_bytes, int local_device,
@@ -432,6 +458,7 @@ static void sync_device(int device) {
 PYBIND11_MODULE(_hipstore, m) {
   m.doc() = "torchstore_amd native core (HIP/CDNA4, gfx950)";
   m.def("ipc_export", &ipc_export, py::arg("ptr"), py::arg("device"));
+  m.def("ipc_export_cache_clear", &ipc_export_cache_clear);
   m.def("ipc_open", &ipc_open, py::arg("handle"), py::arg("local_device"),
         py::arg("src_device"));
   m.def("ipc_close", &ipc_close, py::arg("base"), py::arg("local_device"));
