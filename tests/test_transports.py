@@ -1,0 +1,56 @@
+"""Transport-buffer unit tests (no actors, CPU only)."""
+
+import torch
+
+from torchstore_amd.runtime import serde
+from torchstore_amd.transport.hip_ipc import HipIpcTransportBuffer, IpcDescriptor
+from torchstore_amd.transport.shm import ShmDescriptor, ShmTransportBuffer
+from torchstore_amd.transport.rpc import RpcTransportBuffer
+from torchstore_amd.transport.base import TransportContext
+
+
+def test_ipc_buffer_strips_local_tensors():
+    """Staged tensors must never ride the RPC frame (the 1 GB/s bug)."""
+    buf = HipIpcTransportBuffer()
+    buf._hold = [torch.randn(1024), torch.randn(2048)]
+    buf._scratch = {0: torch.randn(512)}
+    buf.payload = [
+        ("ipc", IpcDescriptor(b"h" * 64, 0, 4096, torch.float32, (1024,), 0))
+    ]
+    header, bufs = serde.dumps(buf)
+    assert bufs == [], "local tensors leaked into the serialized buffer"
+    back = serde.loads(header, [])
+    assert back._hold == [] and back._scratch == {}
+    assert back.payload[0][1].nbytes == 4096
+
+
+def test_shm_buffer_serializes_descriptors_only():
+    buf = ShmTransportBuffer()
+    buf.payload = [
+        ("seg", ShmDescriptor(b"/m", b"/n", 64, torch.float32, (16,))),
+        ("obj", {"a": 1}),
+    ]
+    buf.alloc_sizes = [64, None]
+    header, bufs = serde.dumps(buf)
+    assert bufs == []
+    back = serde.loads(header, [])
+    assert back.payload[0][1].name == b"/n"
+    assert back.alloc_sizes == [64, None]
+
+
+def test_rpc_buffer_carries_payload_out_of_band():
+    buf = RpcTransportBuffer()
+    t = torch.randn(256)
+    buf.data = [("tensor", t)]
+    header, bufs = serde.dumps(buf)
+    assert len(bufs) == 1  # the tensor rides as one out-of-band buffer
+    back = serde.loads(header, [bytearray(b) for b in bufs])
+    assert torch.equal(back.data[0][1], t)
+
+
+def test_context_strip_on_serialization():
+    buf = RpcTransportBuffer()
+    buf.bind_client(object(), TransportContext())
+    header, bufs = serde.dumps(buf)
+    back = serde.loads(header, bufs)
+    assert back._client_ctx is None and back._volume_ref is None

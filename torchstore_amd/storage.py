@@ -240,17 +240,36 @@ class StorageVolume(Actor):
 
     @endpoint
     async def put(self, buffer: TransportBuffer, requests: Sequence[Request]):
+        import time
+
+        t0 = time.perf_counter()
         buffer.attach_volume(self.ctx)
         existing = [self.store.find_existing(r) for r in requests]
+        t1 = time.perf_counter()
         values = await buffer.volume_receive(requests, existing, self.device)
+        t2 = time.perf_counter()
         for r, v in zip(requests, values):
             self.store.put(r, v)
+        logger.info(
+            "volume.put n=%d find=%.1fms recv=%.1fms store=%.1fms",
+            len(requests), (t1 - t0) * 1e3, (t2 - t1) * 1e3,
+            (time.perf_counter() - t2) * 1e3,
+        )
 
     @endpoint
     async def get(self, buffer: TransportBuffer, requests: Sequence[Request]):
+        import time
+
+        t0 = time.perf_counter()
         buffer.attach_volume(self.ctx)
         values = [self.store.fetch(r) for r in requests]
-        return await buffer.volume_send(requests, values)
+        t1 = time.perf_counter()
+        out = await buffer.volume_send(requests, values)
+        logger.info(
+            "volume.get n=%d fetch=%.1fms send=%.1fms",
+            len(requests), (t1 - t0) * 1e3, (time.perf_counter() - t1) * 1e3,
+        )
+        return out
 
     @endpoint
     def get_meta(self, requests: Sequence[Request]):

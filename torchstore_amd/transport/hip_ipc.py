@@ -113,6 +113,15 @@ class HipIpcTransportBuffer(TransportBuffer):
         self._hold: List[torch.Tensor] = []       # keep exports alive
         self._scratch: Dict[int, torch.Tensor] = {}  # req idx -> dense scratch
 
+    def __getstate__(self):
+        # local tensor refs must NEVER ride the RPC frame (they would be
+        # silently staged to CPU and serialized — gigabytes over loopback);
+        # same strip as the reference's RdmaContext (monarch_rdma.py:75-78)
+        state = super().__getstate__()
+        state["_hold"] = []
+        state["_scratch"] = {}
+        return state
+
     # ------------------------------------------------------------- put --
     async def client_stage_put(self, requests: Sequence[Request]) -> None:
         payload: List[Tuple[str, Any]] = []
