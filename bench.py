@@ -119,6 +119,16 @@ async def run_bench(args, rank, world, local_rank):
     for _ in range(args.warmup):
         await one_step()
     barrier()
+    # correctness guard (outside the timed region): the norm weight has the
+    # same layout in both placements, so the pulled local shard must equal
+    # the pushed one bit-for-bit
+    probe = "model.norm.weight"
+    src_t = src_sd[probe]
+    dst_t = dst_sd[probe]
+    get_local = getattr(dst_t, "to_local", lambda: dst_t)
+    put_local = getattr(src_t, "to_local", lambda: src_t)
+    if not torch.equal(get_local(), put_local()):
+        raise RuntimeError("bench correctness probe failed: pulled != pushed")
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
