@@ -110,11 +110,23 @@ async def run_bench(args, rank, world, local_rank):
             torch.distributed.barrier()
         torch.cuda.synchronize()
 
-    async def one_step():
-        await ts.put_state_dict(src_sd, "bench")
-        if world > 1:
-            torch.distributed.barrier()  # all shards committed
-        await ts.get_state_dict("bench", dst_sd)
+    if args.mode == "direct":
+
+        async def one_step():
+            # push = handle refresh (registration happens on the first call);
+            # pull = one batched one-sided read into generator memory
+            await ts.put_state_dict(src_sd, "bench", direct=True)
+            if world > 1:
+                torch.distributed.barrier()
+            await ts.get_state_dict("bench", dst_sd, direct=True)
+
+    else:
+
+        async def one_step():
+            await ts.put_state_dict(src_sd, "bench")
+            if world > 1:
+                torch.distributed.barrier()  # all shards committed
+            await ts.get_state_dict("bench", dst_sd)
 
     for _ in range(args.warmup):
         await one_step()
@@ -143,7 +155,10 @@ async def run_bench(args, rank, world, local_rank):
         elapsed = t.item()
 
     ms_per_step = elapsed / args.steps * 1e3
-    moved = 2 * payload_bytes  # put + get per step, whole job
+    if args.mode == "direct":
+        moved = payload_bytes  # one one-sided read of the model per step
+    else:
+        moved = 2 * payload_bytes  # put + get per step, whole job
     gbps = moved / (elapsed / args.steps) / 1e9
 
     if rank == 0:

@@ -36,6 +36,18 @@ def main():
     t = torch.randn(1 << 20, device="cuda")
     timeit("ipc_export same tensor", lambda: e.ipc_export(t.data_ptr(), 0))
 
+    # can a process open its OWN handle? (decides bench --mode direct shape)
+    try:
+        h, off = e.ipc_export(t.data_ptr(), 0)
+        base = e.ipc_open(bytes(h), 0, 0)
+        probe = torch.empty_like(t)
+        e.copy_batch([(probe.data_ptr(), 0, base + off, 0, t.numel() * 4)])
+        ok = torch.equal(probe, t)
+        print(f"self ipc_open: OK, data match={ok}")
+        e.ipc_close(base, 0)
+    except Exception as exc:
+        print(f"self ipc_open: FAILED ({exc})")
+
     ts = [torch.randn(1 << 18, device="cuda") for _ in range(64)]
 
     def export_many():

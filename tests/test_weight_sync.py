@@ -1,0 +1,199 @@
+"""Direct weight sync plan/pull logic on CPU via the fake memory codec
+(mirrors the reference's MockRDMABuffer test strategy)."""
+
+import pytest
+import torch
+
+import torchstore_amd as ts
+from torchstore_amd.strategy import SingletonStrategy
+from torchstore_amd.transport import TransportType
+from torchstore_amd.types import TensorSlice
+from torchstore_amd import weight_sync
+from torchstore_amd.weight_sync import (
+    DirectWeightSyncDest,
+    DirectWeightSyncSource,
+    FakeMemoryCodec,
+    WeightHandle,
+)
+
+
+@pytest.fixture
+def fake_codec():
+    codec = FakeMemoryCodec()
+    weight_sync.set_codec(codec)
+    yield codec
+    weight_sync.set_codec(None)
+
+
+async def _with_store(body):
+    await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(transport=TransportType.RPC),
+        storage_device="cpu",
+    )
+    try:
+        await body()
+    finally:
+        await ts.shutdown()
+
+
+async def test_exact_match_zero_copy_and_liveness(fake_codec):
+    async def body():
+        c = ts.client()
+        w = torch.randn(64, 64)
+        b = torch.randn(64)
+        src = DirectWeightSyncSource(c, "sync", rank=0, world_size=1)
+        await src.push({"m": {"w": w, "b": b}})
+
+        dst_sd = {"m": {"w": torch.zeros(64, 64), "b": torch.zeros(64)}}
+        dest = DirectWeightSyncDest(c, "sync")
+        await dest.pull(dst_sd)
+        assert torch.equal(dst_sd["m"]["w"], w)
+        assert torch.equal(dst_sd["m"]["b"], b)
+
+        # optimizer-style in-place update is visible WITHOUT re-push
+        with torch.no_grad():
+            w.add_(1.0)
+        await dest.pull(dst_sd)
+        assert torch.equal(dst_sd["m"]["w"], w)
+        # plan was cached: second pull did not refetch handles
+        assert fake_codec.read_count == 4
+
+    await _with_store(body)
+
+
+async def test_sharded_to_full_reshard(fake_codec):
+    """Two Shard(0) source shards assemble into a full dest tensor."""
+
+    async def body():
+        c = ts.client()
+        full = torch.arange(64, dtype=torch.float32).reshape(8, 8)
+        handles = []
+        for r in range(2):
+            shard = full[r * 4 : (r + 1) * 4].clone()
+            handles.append(
+                WeightHandle(
+                    name="w",
+                    desc=fake_codec.export(shard),
+                    slice=TensorSlice(
+                        offsets=(r * 4, 0), local_shape=(4, 8),
+                        global_shape=(8, 8), coordinates=(r,), mesh_shape=(2,),
+                    ),
+                )
+            )
+        await c.put("sync/rank_0", handles)
+        await c.put("sync/num_ranks", 1)
+
+        dst = {"w": torch.zeros(8, 8)}
+        dest = DirectWeightSyncDest(c, "sync")
+        await dest.pull(dst)
+        assert torch.equal(dst["w"], full)
+
+    await _with_store(body)
+
+
+async def test_partial_overlap_column_shards(fake_codec):
+    """Column-sharded (TP-style) sources into a full dest: every scatter
+    destination view is strided, exercising the reshard copy path."""
+
+    async def body():
+        c = ts.client()
+        full = torch.randn(8, 8)
+        handles = []
+        for r in range(2):
+            shard = full[:, r * 4 : (r + 1) * 4].clone()
+            handles.append(
+                WeightHandle(
+                    name="w",
+                    desc=fake_codec.export(shard),
+                    slice=TensorSlice(
+                        offsets=(0, r * 4), local_shape=(8, 4),
+                        global_shape=(8, 8), coordinates=(r,), mesh_shape=(2,),
+                    ),
+                )
+            )
+        await c.put("sync/rank_0", handles)
+        await c.put("sync/num_ranks", 1)
+
+        dest = DirectWeightSyncDest(c, "sync")
+        dst = {"w": torch.zeros(8, 8)}
+        await dest.pull(dst)
+        assert torch.equal(dst["w"], full)
+        assert all(op.scatter is not None for op in dest._plan)
+
+    await _with_store(body)
+
+
+async def test_replicated_dedup(fake_codec):
+    async def body():
+        c = ts.client()
+        t = torch.randn(4, 4)
+        handles = []
+        for r in range(3):  # 3 replicas of the same region
+            handles.append(
+                WeightHandle(
+                    name="w",
+                    desc=fake_codec.export(t.clone()),
+                    slice=TensorSlice(
+                        offsets=(0, 0), local_shape=(4, 4), global_shape=(4, 4),
+                        coordinates=(r,), mesh_shape=(3,),
+                    ),
+                )
+            )
+        await c.put("sync/rank_0", handles)
+        await c.put("sync/num_ranks", 1)
+        dest = DirectWeightSyncDest(c, "sync")
+        dst = {"w": torch.zeros(4, 4)}
+        await dest.pull(dst)
+        assert fake_codec.read_count == 1  # deduped to one read
+        assert torch.equal(dst["w"], t)
+
+    await _with_store(body)
+
+
+async def test_transfer_dtype_staging_and_refresh(fake_codec):
+    async def body():
+        c = ts.client()
+        master = torch.randn(32, 32, dtype=torch.float32)
+        src = DirectWeightSyncSource(
+            c, "sync", transfer_dtype=torch.bfloat16, rank=0, world_size=1
+        )
+        await src.push({"w": master})
+
+        dst = {"w": torch.zeros(32, 32, dtype=torch.bfloat16)}
+        dest = DirectWeightSyncDest(c, "sync")
+        await dest.pull(dst)
+        assert torch.equal(dst["w"], master.to(torch.bfloat16))
+
+        # stale until refresh (staging buffer holds the old cast)
+        with torch.no_grad():
+            master.mul_(2.0)
+        await dest.pull(dst)
+        assert not torch.equal(dst["w"], master.to(torch.bfloat16))
+        await src.push({"w": master})  # push == refresh after registration
+        await dest.pull(dst)
+        assert torch.equal(dst["w"], master.to(torch.bfloat16))
+
+    await _with_store(body)
+
+
+async def test_dtype_mismatch_raises(fake_codec):
+    async def body():
+        c = ts.client()
+        src = DirectWeightSyncSource(c, "sync", rank=0, world_size=1)
+        await src.push({"w": torch.randn(8, 8)})
+        dest = DirectWeightSyncDest(c, "sync")
+        with pytest.raises(TypeError):
+            await dest.pull({"w": torch.zeros(8, 8, dtype=torch.bfloat16)})
+
+    await _with_store(body)
+
+
+async def test_missing_source_raises(fake_codec):
+    async def body():
+        c = ts.client()
+        dest = DirectWeightSyncDest(c, "nope")
+        with pytest.raises(RuntimeError, match="no direct weight sync source"):
+            await dest.pull({"w": torch.zeros(2)})
+
+    await _with_store(body)

@@ -79,6 +79,9 @@ def cast_copy(src: torch.Tensor, out: torch.Tensor) -> None:
         raise ValueError("cast_copy numel mismatch")
     if not (src.is_contiguous() and out.is_contiguous()):
         raise ValueError("cast_copy requires contiguous tensors")
+    if src.dtype == out.dtype:
+        out.copy_(src)
+        return
     sc = _DTYPE_CODES.get(src.dtype)
     dc = _DTYPE_CODES.get(out.dtype)
     if (

@@ -65,12 +65,32 @@ def _nbytes(flat: Dict[str, Any]) -> int:
     return total
 
 
+# per-(client, key) direct-sync endpoints, built lazily on first use
+_direct_sources: Dict[tuple, Any] = {}
+_direct_dests: Dict[tuple, Any] = {}
+
+
 async def put_state_dict(
     client: LocalClient,
     state_dict: Dict[str, Any],
     key: str,
     transfer_dtype: Optional[torch.dtype] = None,
+    direct: bool = False,
+    rank: Optional[int] = None,
+    world_size: Optional[int] = None,
 ) -> None:
+    if direct:
+        from torchstore_amd.weight_sync import DirectWeightSyncSource
+
+        ck = (id(client), key)
+        source = _direct_sources.get(ck)
+        if source is None:
+            source = DirectWeightSyncSource(
+                client, key, transfer_dtype, rank=rank, world_size=world_size
+            )
+            _direct_sources[ck] = source
+        await source.push(state_dict)
+        return
     tracker = LatencyTracker(f"put_state_dict[{key}]")
     flat, mapping = _flatten(state_dict)
     tracker.step("flatten")
@@ -89,7 +109,20 @@ async def get_state_dict(
     key: str,
     user_state_dict: Optional[Dict[str, Any]] = None,
     strict: bool = True,
+    direct: bool = False,
 ) -> Dict[str, Any]:
+    if direct:
+        from torchstore_amd.weight_sync import DirectWeightSyncDest
+
+        if user_state_dict is None:
+            raise ValueError("direct get_state_dict needs a destination state_dict")
+        ck = (id(client), key)
+        dest = _direct_dests.get(ck)
+        if dest is None:
+            dest = DirectWeightSyncDest(client, key)
+            _direct_dests[ck] = dest
+        await dest.pull(user_state_dict)
+        return user_state_dict
     tracker = LatencyTracker(f"get_state_dict[{key}]")
     try:
         mapping = await client.get(f"{key}/{MAPPING_KEY}")
