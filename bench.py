@@ -23,6 +23,44 @@ import time
 
 import torch
 
+from torchstore_amd.runtime import Actor, endpoint
+
+
+class GeneratorActor(Actor):
+    """Serving-fleet stand-in for --mode direct: pulls weights one-sided
+    over HIP IPC from the trainer's live parameter memory."""
+
+    def __init__(self, controller, key, layers, world, rank, device_index):
+        import torchstore_amd as ts
+        from torchstore_amd.models import llama
+        from torchstore_amd.strategy import LocalRankStrategy
+        from torchstore_amd.weight_sync import DirectWeightSyncDest
+
+        os.environ["RANK"] = str(rank)
+        torch.cuda.set_device(device_index)
+        ts.attach(controller, LocalRankStrategy())
+        self.dst_sd = llama.make_local_shard_state_dict(
+            rank, world, llama.tp_placement,
+            device=f"cuda:{device_index}", layers=layers,
+        )
+        self.dest = DirectWeightSyncDest(ts.client(), key)
+
+    @endpoint
+    async def pull(self):
+        t0 = time.perf_counter()
+        await self.dest.pull(self.dst_sd)
+        torch.cuda.synchronize()
+        nbytes = sum(op.nbytes for op in self.dest._plan)
+        return time.perf_counter() - t0, nbytes
+
+    @endpoint
+    def verify(self, name, expected):
+        from torchstore_amd.types import LocalShard
+
+        v = self.dst_sd[name]
+        local = v.tensor if isinstance(v, LocalShard) else v
+        return torch.equal(local, expected.to(local.device))
+
 
 def parse_args():
     p = argparse.ArgumentParser()
@@ -99,7 +137,7 @@ async def run_bench(args, rank, world, local_rank):
             controller = pickle.loads(store.get("controller"))
             ts.attach(controller, LocalRankStrategy())
     else:
-        await ts.initialize(
+        controller = await ts.initialize(
             num_storage_volumes=1,
             strategy=LocalRankStrategy(),
             storage_device="auto",
@@ -111,14 +149,27 @@ async def run_bench(args, rank, world, local_rank):
         torch.cuda.synchronize()
 
     if args.mode == "direct":
+        # trainer = this process; generator = a separate process on the same
+        # GPU (IPC handles cannot be opened by their exporting process)
+        from torchstore_amd.runtime import spawn_actors
+
+        await ts.put_state_dict(
+            src_sd, "bench", direct=True, rank=rank, world_size=world
+        )
+        if world > 1:
+            torch.distributed.barrier()
+        gen_mesh = await asyncio.to_thread(
+            spawn_actors, 1, GeneratorActor, f"generator-{rank}",
+            controller, "bench", layers, world, rank, local_rank,
+        )
+        gen = gen_mesh.handles[0]
+        pull_bytes = [0]
 
         async def one_step():
-            # push = handle refresh (registration happens on the first call);
+            # push = staging refresh (no cast here → sync only);
             # pull = one batched one-sided read into generator memory
             await ts.put_state_dict(src_sd, "bench", direct=True)
-            if world > 1:
-                torch.distributed.barrier()
-            await ts.get_state_dict("bench", dst_sd, direct=True)
+            _dt, pull_bytes[0] = await gen.pull.call_one()
 
     else:
 
@@ -136,11 +187,16 @@ async def run_bench(args, rank, world, local_rank):
     # the pushed one bit-for-bit
     probe = "model.norm.weight"
     src_t = src_sd[probe]
-    dst_t = dst_sd[probe]
-    get_local = getattr(dst_t, "to_local", lambda: dst_t)
     put_local = getattr(src_t, "to_local", lambda: src_t)
-    if not torch.equal(get_local(), put_local()):
-        raise RuntimeError("bench correctness probe failed: pulled != pushed")
+    if args.mode == "direct":
+        ok = await gen.verify.call_one(probe, put_local().cpu())
+        if not ok:
+            raise RuntimeError("bench correctness probe failed (direct pull)")
+    else:
+        dst_t = dst_sd[probe]
+        get_local = getattr(dst_t, "to_local", lambda: dst_t)
+        if not torch.equal(get_local(), put_local()):
+            raise RuntimeError("bench correctness probe failed: pulled != pushed")
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
@@ -156,7 +212,11 @@ async def run_bench(args, rank, world, local_rank):
 
     ms_per_step = elapsed / args.steps * 1e3
     if args.mode == "direct":
-        moved = payload_bytes  # one one-sided read of the model per step
+        moved = pull_bytes[0]  # actual bytes read per pull, this rank
+        if world > 1:
+            t = torch.tensor([float(moved)], device=device)
+            torch.distributed.all_reduce(t)
+            moved = int(t.item())
     else:
         moved = 2 * payload_bytes  # put + get per step, whole job
     gbps = moved / (elapsed / args.steps) / 1e9
@@ -185,6 +245,8 @@ async def run_bench(args, rank, world, local_rank):
             },
         }))
 
+    if args.mode == "direct":
+        await gen_mesh.stop()
     if world > 1:
         torch.distributed.barrier()
     if rank == 0 or world == 1:

@@ -119,7 +119,9 @@ async def test_partial_overlap_column_shards(fake_codec):
         dst = {"w": torch.zeros(8, 8)}
         await dest.pull(dst)
         assert torch.equal(dst["w"], full)
-        assert all(op.scatter is not None for op in dest._plan)
+        # column shards land via pitched reads of only the overlap bytes
+        assert all(op.kind == "2d" for op in dest._plan)
+        assert sum(op.nbytes for op in dest._plan) == full.numel() * 4
 
     await _with_store(body)
 

@@ -215,6 +215,10 @@ class LocalClient:
         """Destination local tensor + wanted global region from a `like`."""
         if like is None:
             return None, None
+        from torchstore_amd.types import LocalShard
+
+        if isinstance(like, LocalShard):
+            return like.tensor, like.slice
         from torch.distributed.tensor import DTensor
 
         if isinstance(like, DTensor):

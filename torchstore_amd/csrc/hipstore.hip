@@ -44,6 +44,8 @@ namespace py = pybind11;
   do {                                                                         \
     hipError_t _e = (expr);                                                    \
     if (_e != hipSuccess) {                                                    \
+      (void)hipGetLastError(); /* clear the sticky error so the caller's */    \
+      /* process (torch's own error checks) is not poisoned */                 \
       throw std::runtime_error(std::string(#expr) + " failed: " +              \
                                hipGetErrorString(_e));                         \
     }                                                                          \
@@ -184,6 +186,35 @@ static void copy_batch(
                                    reinterpret_cast<void*>(src), src_dev, n,
                                    s));
     }
+    used.push_back(s);
+  }
+  for (hipStream_t s : used) HIP_CHECK(hipStreamSynchronize(s));
+}
+
+// copies: (dst_ptr, dst_dev, dpitch, src_ptr, src_dev, spitch, width, height)
+// — strided (pitched) one-sided reads/writes: moves ONLY the overlap bytes
+// of a reshard instead of whole remote shards (the reference always reads
+// the full source shard, direct_weight_sync.py:280-314).
+static void copy_batch_2d(
+    const std::vector<std::tuple<uintptr_t, int, uint64_t, uintptr_t, int,
+                                 uint64_t, uint64_t, uint64_t>>& copies) {
+  if (copies.empty()) return;
+  std::vector<hipStream_t> used;
+  int i = 0;
+  for (const auto& c : copies) {
+    uintptr_t dst = std::get<0>(c);
+    int dst_dev = std::get<1>(c);
+    uint64_t dpitch = std::get<2>(c);
+    uintptr_t src = std::get<3>(c);
+    uint64_t spitch = std::get<5>(c);
+    uint64_t width = std::get<6>(c);
+    uint64_t height = std::get<7>(c);
+    DevicePool& p = pool_for(dst_dev);
+    hipStream_t s = p.streams[i++ % kStreamsPerDevice];
+    HIP_CHECK(hipSetDevice(dst_dev));
+    HIP_CHECK(hipMemcpy2DAsync(reinterpret_cast<void*>(dst), dpitch,
+                               reinterpret_cast<void*>(src), spitch, width,
+                               height, hipMemcpyDeviceToDevice, s));
     used.push_back(s);
   }
   for (hipStream_t s : used) HIP_CHECK(hipStreamSynchronize(s));
@@ -476,6 +507,8 @@ PYBIND11_MODULE(_hipstore, m) {
         py::arg("src_device"));
   m.def("ipc_close", &ipc_close, py::arg("base"), py::arg("local_device"));
   m.def("copy_batch", &copy_batch, py::arg("copies"),
+        py::call_guard<py::gil_scoped_release>());
+  m.def("copy_batch_2d", &copy_batch_2d, py::arg("copies"),
         py::call_guard<py::gil_scoped_release>());
   m.def("copy_slices", &copy_slices, py::arg("slices"), py::arg("device"),
         py::arg("stream"), py::arg("blocking") = true,

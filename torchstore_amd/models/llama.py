@@ -65,6 +65,58 @@ def tp_placement(shape: Tuple[int, ...], world: int) -> Optional[int]:
     return None
 
 
+def make_local_shard_state_dict(
+    rank: int,
+    world: int,
+    shard_fn,
+    dtype: torch.dtype = torch.bfloat16,
+    device: str = "cuda",
+    layers: int = LAYERS,
+    zero: bool = True,
+):
+    """DTensor-free sharded state_dict: {name: LocalShard | tensor}.
+
+    For processes outside any torch.distributed world (serving fleets).
+    Shard dims chosen by ``shard_fn`` are always divisible by ``world``
+    for these shapes, so the even split matches DTensor chunking exactly.
+    """
+    from torchstore_amd.types import LocalShard, TensorSlice
+
+    shapes = llama3_8b_shapes(layers)
+    out = {}
+    for name, shape in shapes.items():
+        if world == 1:
+            t = torch.empty(shape, dtype=dtype, device=device)
+            if not zero:
+                t.normal_(0, 0.02)
+            out[name] = t
+            continue
+        dim = shard_fn(shape, world)
+        if dim is None:
+            local_shape = shape
+            offsets = (0,) * len(shape)
+        else:
+            assert shape[dim] % world == 0
+            local_shape = list(shape)
+            local_shape[dim] = shape[dim] // world
+            local_shape = tuple(local_shape)
+            offsets = tuple(
+                rank * (shape[d] // world) if d == dim else 0
+                for d in range(len(shape))
+            )
+        t = torch.empty(local_shape, dtype=dtype, device=device)
+        if not zero:
+            t.normal_(0, 0.02)
+        out[name] = LocalShard(
+            tensor=t,
+            slice=TensorSlice(
+                offsets=offsets, local_shape=local_shape, global_shape=shape,
+                coordinates=(rank,), mesh_shape=(world,),
+            ),
+        )
+    return out
+
+
 def make_sharded_state_dict(
     mesh,
     shard_fn,

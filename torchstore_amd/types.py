@@ -70,6 +70,18 @@ class TensorSlice:
         return itertools.product(*(range(s) for s in mesh_shape))
 
 
+@dataclass
+class LocalShard:
+    """A shard + its placement, without requiring a live DTensor/mesh.
+
+    Lets processes outside any torch.distributed world (e.g. a serving
+    fleet pulling weights) name sharded layouts explicitly.
+    """
+
+    tensor: "torch.Tensor"
+    slice: TensorSlice
+
+
 def _dtensor_is_trivially_local(value) -> bool:
     """A DTensor whose local tensor IS the full tensor on every rank."""
     from torch.distributed.tensor.placement_types import Replicate
@@ -133,6 +145,8 @@ class Request:
                 tensor_val=value.to_local(),
                 tensor_slice=slice_from_dtensor(value),
             )
+        if isinstance(value, LocalShard):
+            return cls(key=key, tensor_val=value.tensor, tensor_slice=value.slice)
         if isinstance(value, torch.Tensor):
             return cls(key=key, tensor_val=value)
         if value is None:

@@ -55,29 +55,72 @@ class WeightHandle:
 # ---------------------------------------------------------------------------
 
 
+@dataclass
+class _TransferOp:
+    """One one-sided read of a pull.
+
+    kinds:
+      * ``1d``   — whole descriptor straight into a contiguous dest (the
+        zero-copy exact-match case: dest is live parameter memory);
+      * ``2d``   — pitched read of ONLY the overlap region straight into a
+        (possibly strided) dest view — moves overlap bytes, not whole
+        shards (improvement over the reference's full-shard reads);
+      * ``recv`` — full shard into a cached recv buffer + a K2 scatter
+        (fallback for >2-D overlap geometry).
+    """
+
+    kind: str
+    desc: IpcDescriptor
+    dst: torch.Tensor
+    src_offset_bytes: int = 0
+    spitch: int = 0          # remote row pitch (bytes), 2d only
+    width: int = 0           # row bytes, 2d only
+    height: int = 0          # rows, 2d only
+    scatter: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
+
+    @property
+    def nbytes(self) -> int:
+        if self.kind == "2d":
+            return self.width * self.height
+        return self.desc.nbytes
+
+
 class IpcMemoryCodec:
     def export(self, t: torch.Tensor) -> IpcDescriptor:
         from torchstore_amd.transport.hip_ipc import export_tensor
 
         return export_tensor(t)
 
-    def read_batch(
-        self, ops: Sequence[Tuple[torch.Tensor, IpcDescriptor]], ctx
-    ) -> None:
-        """Read each descriptor's bytes into the paired local tensor."""
+    def read_batch(self, ops: Sequence[_TransferOp], ctx) -> None:
+        """Execute every read as one striped batch over the stream pool."""
         from torchstore_amd.transport.hip_ipc import IpcOpenCache
         from torchstore_amd.ops import gpu
 
         cache: IpcOpenCache = ctx.cache(IpcOpenCache)
-        copies = []
-        for dst, desc in ops:
-            assert dst.is_contiguous()
-            src_ptr = cache.resolve(desc, dst.device.index)
-            copies.append(
-                (dst.data_ptr(), dst.device.index, src_ptr, desc.device_index,
-                 desc.nbytes)
-            )
-        gpu.copy_batch(copies)
+        copies_1d = []
+        copies_2d = []
+        for op in ops:
+            dev = op.dst.device.index
+            src_ptr = cache.resolve(op.desc, dev)
+            if op.kind == "2d":
+                es = op.dst.element_size()
+                dpitch = (
+                    op.dst.stride(0) * es if op.dst.dim() == 2 else op.width
+                )
+                copies_2d.append(
+                    (op.dst.data_ptr(), dev, dpitch,
+                     src_ptr + op.src_offset_bytes, op.desc.device_index,
+                     op.spitch, op.width, op.height)
+                )
+            else:
+                copies_1d.append(
+                    (op.dst.data_ptr(), dev, src_ptr, op.desc.device_index,
+                     op.desc.nbytes)
+                )
+        if copies_1d:
+            gpu.copy_batch(copies_1d)
+        if copies_2d:
+            gpu.ext().copy_batch_2d(copies_2d)
 
 
 class FakeMemoryCodec:
@@ -100,10 +143,21 @@ class FakeMemoryCodec:
             device_index=-1,
         )
 
-    def read_batch(self, ops, ctx) -> None:
-        for dst, desc in ops:
-            src = self.registry[int.from_bytes(desc.handle, "little")]
-            byte_view(dst)[:].copy_(byte_view(src.contiguous()))
+    def read_batch(self, ops: Sequence[_TransferOp], ctx) -> None:
+        for op in ops:
+            src = self.registry[int.from_bytes(op.desc.handle, "little")]
+            src_c = src.contiguous()
+            if op.kind == "2d":
+                es = src_c.element_size()
+                assert op.src_offset_bytes % es == 0 and op.spitch % es == 0
+                view = src_c.reshape(-1).as_strided(
+                    (op.height, op.width // es),
+                    (op.spitch // es, 1),
+                    storage_offset=op.src_offset_bytes // es,
+                )
+                op.dst.copy_(view.reshape(op.dst.shape))
+            else:
+                byte_view(op.dst)[:].copy_(byte_view(src_c))
             self.read_count += 1
 
 
@@ -217,14 +271,6 @@ class DirectWeightSyncSource:
             torch.cuda.synchronize()
 
 
-@dataclass
-class _TransferOp:
-    dst: torch.Tensor            # contiguous read target (param or recv buf)
-    desc: IpcDescriptor
-    # set for partial overlaps: scatter recv->dest after the read
-    scatter: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
-
-
 class DirectWeightSyncDest:
     """Generator side: cached plan, one batched read per pull."""
 
@@ -277,6 +323,11 @@ class DirectWeightSyncDest:
                 if region_key in covered:
                     continue  # replicated shard — read once
                 covered.add(region_key)
+                if dest_local.dtype != h.desc.dtype:
+                    raise TypeError(
+                        f"{name}: dest dtype {dest_local.dtype} != "
+                        f"transfer dtype {h.desc.dtype}"
+                    )
                 exact = (
                     inter.offsets == wanted.offsets
                     and inter.local_shape == wanted.local_shape
@@ -284,25 +335,49 @@ class DirectWeightSyncDest:
                     and inter.local_shape == src_region.local_shape
                 )
                 if exact and dest_local.is_contiguous():
-                    if dest_local.dtype != h.desc.dtype:
-                        raise TypeError(
-                            f"{name}: dest dtype {dest_local.dtype} != "
-                            f"transfer dtype {h.desc.dtype}"
-                        )
-                    plan.append(_TransferOp(dst=dest_local, desc=h.desc))
+                    plan.append(_TransferOp(kind="1d", dst=dest_local, desc=h.desc))
                     continue
-                # partial overlap: full-shard read + sliced scatter
+                dst_view = region_view(
+                    dest_local, wanted.offsets, inter.offsets, inter.local_shape
+                )
+                es = dest_local.element_size()
+                shape = inter.local_shape
+                rel = tuple(
+                    inter.offsets[d] - src_region.offsets[d]
+                    for d in range(len(shape))
+                )
+                if len(shape) <= 2 and (
+                    dst_view.dim() == 0 or dst_view.stride(-1) == 1
+                ):
+                    # pitched read of exactly the overlap bytes
+                    width = shape[-1] * es if shape else es
+                    height = shape[0] if len(shape) == 2 else 1
+                    if len(h.desc.shape) == 2:
+                        spitch = h.desc.shape[1] * es
+                        src_off = (rel[0] * h.desc.shape[1] + rel[-1]) * es
+                    else:
+                        spitch = width
+                        src_off = rel[0] * es if rel else 0
+                    plan.append(
+                        _TransferOp(
+                            kind="2d", dst=dst_view, desc=h.desc,
+                            src_offset_bytes=src_off, spitch=spitch,
+                            width=width, height=height,
+                        )
+                    )
+                    continue
+                # >2-D geometry: full-shard read + batched K2 scatter
                 recv = torch.empty(
                     h.desc.shape, dtype=h.desc.dtype, device=dest_local.device
                 )
                 src_view = region_view(
                     recv, src_region.offsets, inter.offsets, inter.local_shape
                 )
-                dst_view = region_view(
-                    dest_local, wanted.offsets, inter.offsets, inter.local_shape
-                )
                 plan.append(
-                    _TransferOp(dst=recv, desc=h.desc, scatter=(src_view, dst_view))
+                    _TransferOp(
+                        kind="recv", dst=recv, desc=h.desc,
+                        scatter=(src_view, dst_view),
+                    )
                 )
         return plan
 
@@ -313,14 +388,16 @@ class DirectWeightSyncDest:
             dest_flat, _ = _flatten(dest_state_dict)
             self._plan = self._build_plan(handles, dest_flat)
             logger.info(
-                "direct sync plan: %d ops (%d resharded)",
+                "direct sync plan: %d ops (%d zero-copy, %d pitched, %d recv)",
                 len(self._plan),
-                sum(1 for op in self._plan if op.scatter),
+                sum(1 for op in self._plan if op.kind == "1d"),
+                sum(1 for op in self._plan if op.kind == "2d"),
+                sum(1 for op in self._plan if op.kind == "recv"),
             )
         tracker = LatencyTracker(f"direct_sync.pull[{self.key}]")
-        ops = [(op.dst, op.desc) for op in self._plan]
-        codec.read_batch(ops, self.client._ctx)
-        tracker.step("read", sum(op.desc.nbytes for op in self._plan))
+        codec.read_batch(self._plan, self.client._ctx)
+        nbytes = sum(op.nbytes for op in self._plan)
+        tracker.step("read", nbytes)
         scatters = [op.scatter for op in self._plan if op.scatter is not None]
         if scatters:
             device = scatters[0][1].device
@@ -331,6 +408,5 @@ class DirectWeightSyncDest:
             else:
                 for src_view, dst_view in scatters:
                     dst_view.copy_(src_view)
-        nbytes = sum(op.desc.nbytes for op in self._plan)
         tracker.step("scatter")
         tracker.e2e(nbytes)
