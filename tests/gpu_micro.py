@@ -96,6 +96,30 @@ def main():
     )
     print(f"  bw (rd+wr): {nb/dt/1e9:.1f} GB/s")
 
+    # K2 batched many-slice scatter vs a torch copy loop (the reshard case)
+    n_slices = 256
+    srcs2 = [torch.randn(512, 512, device="cuda", dtype=torch.bfloat16)
+             for _ in range(n_slices)]
+    dest_big = torch.zeros(512, 512 * n_slices, device="cuda", dtype=torch.bfloat16)
+    pairs = [
+        (srcs2[i], dest_big[:, i * 512 : (i + 1) * 512]) for i in range(n_slices)
+    ]
+    nb2 = sum(s.numel() * 2 for s in srcs2) * 2
+    dt = timeit(
+        f"copy_pairs {n_slices}x512x512 bf16 scatter",
+        lambda: gpu.copy_pairs(pairs, dest_big.device, blocking=True),
+        n=10,
+    )
+    print(f"  bw (rd+wr): {nb2/dt/1e9:.1f} GB/s")
+
+    def torch_loop():
+        for s, d in pairs:
+            d.copy_(s)
+        torch.cuda.synchronize()
+
+    dt = timeit("torch copy_ loop same scatter", torch_loop, n=10)
+    print(f"  bw (rd+wr): {nb2/dt/1e9:.1f} GB/s")
+
     # K3 cast bandwidth
     f = torch.randn(1 << 28, device="cuda")  # 1 GiB fp32
     o = torch.empty(1 << 28, dtype=torch.bfloat16, device="cuda")

@@ -259,10 +259,18 @@ __device__ __forceinline__ void row_offsets(const SliceDesc& d, uint64_t row,
   }
 }
 
+// One WAVE per work unit (a <=16 KiB span of one row): a 256-thread block
+// runs 4 independent waves, so rows never serialize behind each other
+// inside a block, and the x4-unrolled 16B path keeps 4 loads in flight
+// per lane (64 lanes x 16B x 4 = 4 KiB outstanding per wave).
 __global__ void __launch_bounds__(256)
 copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
                    uint64_t total_units) {
-  for (uint64_t unit = blockIdx.x; unit < total_units; unit += gridDim.x) {
+  const uint32_t wave = threadIdx.x >> 6;
+  const uint32_t lane = threadIdx.x & 63u;
+  const uint64_t stride = (uint64_t)gridDim.x * 4;
+  for (uint64_t unit = (uint64_t)blockIdx.x * 4 + wave; unit < total_units;
+       unit += stride) {
     // binary search: greatest s with units_prefix <= unit
     uint32_t lo = 0, hi = nslices - 1;
     while (lo < hi) {
@@ -282,36 +290,33 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
     uintptr_t sa = reinterpret_cast<uintptr_t>(src);
     uintptr_t da = reinterpret_cast<uintptr_t>(dst);
     if (((sa | da | len) & 15u) == 0) {
-      // 16B vector path: coalesced dwordx4. Unrolled x4 so four loads are
-      // in flight before the first dependent store (ILP hides HBM latency).
       const uint4* s4 = reinterpret_cast<const uint4*>(src);
       uint4* d4 = reinterpret_cast<uint4*>(dst);
       uint32_t n4 = len >> 4;
-      uint32_t bd = blockDim.x;
-      uint32_t i = threadIdx.x;
-      for (; i + 3 * bd < n4; i += 4 * bd) {
+      uint32_t i = lane;
+      for (; i + 192 < n4; i += 256) {
         uint4 a = s4[i];
-        uint4 b = s4[i + bd];
-        uint4 c = s4[i + 2 * bd];
-        uint4 d = s4[i + 3 * bd];
+        uint4 b = s4[i + 64];
+        uint4 c = s4[i + 128];
+        uint4 e = s4[i + 192];
         d4[i] = a;
-        d4[i + bd] = b;
-        d4[i + 2 * bd] = c;
-        d4[i + 3 * bd] = d;
+        d4[i + 64] = b;
+        d4[i + 128] = c;
+        d4[i + 192] = e;
       }
-      for (; i < n4; i += bd) d4[i] = s4[i];
+      for (; i < n4; i += 64) d4[i] = s4[i];
     } else if (((sa | da | len) & 3u) == 0) {
       const uint32_t* s1 = reinterpret_cast<const uint32_t*>(src);
       uint32_t* d1 = reinterpret_cast<uint32_t*>(dst);
       uint32_t n1 = len >> 2;
-      for (uint32_t i = threadIdx.x; i < n1; i += blockDim.x) d1[i] = s1[i];
+      for (uint32_t i = lane; i < n1; i += 64) d1[i] = s1[i];
     } else if (((sa | da | len) & 1u) == 0) {
       const uint16_t* s1 = reinterpret_cast<const uint16_t*>(src);
       uint16_t* d1 = reinterpret_cast<uint16_t*>(dst);
       uint32_t n1 = len >> 1;
-      for (uint32_t i = threadIdx.x; i < n1; i += blockDim.x) d1[i] = s1[i];
+      for (uint32_t i = lane; i < n1; i += 64) d1[i] = s1[i];
     } else {
-      for (uint32_t i = threadIdx.x; i < len; i += blockDim.x) dst[i] = src[i];
+      for (uint32_t i = lane; i < len; i += 64) dst[i] = src[i];
     }
   }
 }
@@ -373,8 +378,9 @@ static void copy_slices(const std::vector<PySlice>& slices, int device,
     HIP_CHECK(hipMemcpyAsync(p.d_desc, p.h_desc, bytes, hipMemcpyHostToDevice,
                              stream));
     HIP_CHECK(hipEventRecord(p.desc_evt, stream));
-    // memory-bound: cap grid at 2048 blocks, grid-stride the rest (guide G11)
-    grid = (uint32_t)std::min<uint64_t>(units, 2048);
+    // memory-bound: cap grid at 2048 blocks, grid-stride the rest (guide
+    // G11); each block consumes 4 units (one per wave)
+    grid = (uint32_t)std::min<uint64_t>((units + 3) / 4, 2048);
     hipLaunchKernelGGL(copy_slices_kernel, dim3(grid), dim3(256), 0, stream,
                        reinterpret_cast<const SliceDesc*>(p.d_desc),
                        (uint32_t)n, units);
