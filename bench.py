@@ -113,29 +113,12 @@ async def run_bench(args, rank, world, local_rank):
     shapes = llama.llama3_8b_shapes(layers)
     payload_bytes = llama.total_bytes(shapes, torch.bfloat16)
 
-    # store bring-up: rank 0 spawns volumes (one per GPU) + controller and
-    # publishes the controller handle through a TCPStore
+    # store bring-up: rank 0 spawns volumes (one per GPU) + controller; the
+    # SPMD bootstrap shares the controller handle through its rendezvous
     if world > 1:
-        from torch.distributed import TCPStore
-
-        store = TCPStore(
-            os.environ["MASTER_ADDR"], int(os.environ["MASTER_PORT"]) + 1,
-            world, is_master=rank == 0,
+        controller = await ts.initialize_spmd(
+            strategy=LocalRankStrategy(), storage_device="auto"
         )
-        if rank == 0:
-            controller = await ts.initialize(
-                num_storage_volumes=world,
-                strategy=LocalRankStrategy(),
-                storage_device="auto",
-            )
-            import pickle
-
-            store.set("controller", pickle.dumps(controller))
-        else:
-            import pickle
-
-            controller = pickle.loads(store.get("controller"))
-            ts.attach(controller, LocalRankStrategy())
     else:
         controller = await ts.initialize(
             num_storage_volumes=1,
@@ -249,8 +232,7 @@ async def run_bench(args, rank, world, local_rank):
         await gen_mesh.stop()
     if world > 1:
         torch.distributed.barrier()
-    if rank == 0 or world == 1:
-        await ts.shutdown()
+    await ts.shutdown()  # collective in SPMD mode
     if world > 1:
         torch.distributed.destroy_process_group()
 

@@ -31,6 +31,7 @@ from torchstore_amd.api import (
     reset_client,
     shutdown,
 )
+from torchstore_amd.spmd import SPMDEnv, initialize_spmd, shutdown_spmd
 from torchstore_amd.strategy import (
     HostStrategy,
     LocalRankStrategy,
@@ -61,6 +62,9 @@ __all__ = [
     "put_state_dict",
     "reset_client",
     "shutdown",
+    "SPMDEnv",
+    "initialize_spmd",
+    "shutdown_spmd",
     "HostStrategy",
     "LocalRankStrategy",
     "PlacementStrategy",

@@ -122,6 +122,14 @@ def client(store_name: str = DEFAULT_STORE) -> LocalClient:
 
 
 async def shutdown(store_name: str = DEFAULT_STORE) -> None:
+    # SPMD jobs route through the collective session (rank-0 teardown +
+    # status broadcast) when one exists
+    from torchstore_amd import spmd as _spmd
+
+    spmd_session = _spmd._spmd_sessions.pop(store_name, None)
+    if spmd_session is not None:
+        await spmd_session.shutdown()
+        return
     session = _sessions.pop(store_name, None)
     c = _clients.pop(store_name, None)
     if c is not None:
