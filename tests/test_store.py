@@ -11,7 +11,13 @@ from torchstore_amd.strategy import SingletonStrategy
 from torchstore_amd.transport import TransportType
 
 
-@pytest.fixture(params=[TransportType.RPC, TransportType.SHARED_MEMORY])
+@pytest.fixture(
+    params=[
+        TransportType.RPC,
+        TransportType.SHARED_MEMORY,
+        TransportType.GLOO,
+    ]
+)
 def transport(request):
     return request.param
 

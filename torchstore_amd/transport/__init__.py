@@ -18,6 +18,7 @@ from __future__ import annotations
 from typing import Optional
 
 import torch
+import torch.distributed
 
 from torchstore_amd.transport.base import (
     TransportBuffer,
@@ -42,10 +43,16 @@ __all__ = [
 def _registry():
     from torchstore_amd.transport.shm import ShmTransportBuffer
     from torchstore_amd.transport.hip_ipc import HipIpcTransportBuffer
+    from torchstore_amd.transport.pg import (
+        GlooTransportBuffer,
+        RcclTransportBuffer,
+    )
     return {
         TransportType.RPC: RpcTransportBuffer,
         TransportType.SHARED_MEMORY: ShmTransportBuffer,
         TransportType.HIP_IPC: HipIpcTransportBuffer,
+        TransportType.GLOO: GlooTransportBuffer,
+        TransportType.RCCL: RcclTransportBuffer,
     }
 
 
@@ -72,6 +79,15 @@ def resolve_transport_type(volume_ref) -> TransportType:
         return TransportType.HIP_IPC
     if _shm_available(volume_ref):
         return TransportType.SHARED_MEMORY
+    if not volume_ref.is_local:
+        if (
+            _env_on("TORCHSTORE_AMD_RCCL_ENABLED")
+            and torch.cuda.is_available()
+            and volume_ref.device.startswith("cuda")
+        ):
+            return TransportType.RCCL
+        if _env_on("TORCHSTORE_AMD_GLOO_ENABLED") and torch.distributed.is_gloo_available():
+            return TransportType.GLOO
     return TransportType.RPC
 
 
