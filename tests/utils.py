@@ -141,6 +141,41 @@ class DTensorWorker(Actor):
             raise AssertionError(f"object entry mismatch: {out['model']['step']}")
         return "ok"
 
+    @endpoint
+    async def dcp_save_and_push(
+        self, ckpt_path: str, sd_key: str, shape, mesh_shape, placements
+    ):
+        import torch.distributed.checkpoint as dcp
+
+        full = make_full_tensor(shape)
+        dt = self._dtensor(full, mesh_shape, placements)
+        sd = {"w": dt, "meta": 42}
+        dcp.save({"w": dt}, checkpoint_id=ckpt_path)
+        await api.put_state_dict(sd, sd_key)
+        return "ok"
+
+    @endpoint
+    async def dcp_load_and_compare(
+        self, ckpt_path: str, sd_key: str, shape, mesh_shape, placements
+    ):
+        """Ground truth: DCP's own resharding load must equal our pull."""
+        import torch.distributed.checkpoint as dcp
+
+        full = make_full_tensor(shape)
+        dcp_dest = self._dtensor(torch.zeros_like(full), mesh_shape, placements)
+        dcp_sd = {"w": dcp_dest}
+        dcp.load(dcp_sd, checkpoint_id=ckpt_path)
+
+        ts_dest = self._dtensor(torch.zeros_like(full), mesh_shape, placements)
+        out = await api.get_state_dict(sd_key, {"w": ts_dest, "meta": 0})
+        if not torch.equal(out["w"].to_local(), dcp_sd["w"].to_local()):
+            raise AssertionError(
+                f"rank {self.rank}: torchstore reshard != DCP reshard"
+            )
+        if out["meta"] != 42:
+            raise AssertionError("object entry lost")
+        return "ok"
+
     def teardown_local(self):
         if dist.is_initialized():
             dist.destroy_process_group()
