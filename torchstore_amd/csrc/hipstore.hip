@@ -107,6 +107,10 @@ static void ensure_desc_capacity(DevicePool& p, int device, size_t bytes) {
 static std::mutex g_export_mutex;
 static std::unordered_map<uintptr_t, std::string> g_export_cache;
 
+// Returns (handle_bytes, offset_in_block, block_size).  Callers MUST route
+// blocks >= 2 GiB through the chunked-staging path: hipIpcOpenMemHandle of
+// a >=2^31-byte dmabuf hangs on this platform (measured; the export itself
+// succeeds, the peer's import never returns).
 static py::tuple ipc_export(uintptr_t ptr, int device) {
   HIP_CHECK(hipSetDevice(device));
   void* base = nullptr;
@@ -129,7 +133,8 @@ static py::tuple ipc_export(uintptr_t ptr, int device) {
     g_export_cache.emplace(base_u, handle_str);
   }
   return py::make_tuple(py::bytes(handle_str),
-                        static_cast<uint64_t>(ptr - base_u));
+                        static_cast<uint64_t>(ptr - base_u),
+                        static_cast<uint64_t>(size));
 }
 
 static void ipc_export_cache_clear() {

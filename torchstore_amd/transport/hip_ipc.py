@@ -4,13 +4,13 @@ This is the MI355X replacement for the reference's ibverbs RDMA transports
 (torchstore ``transport/monarch_rdma.py`` / ``torchcomms``): one-sided bulk
 byte movement with no staging hop.
 
-Mechanics (native side in ``csrc/ipc.cpp``):
+Mechanics (native side in ``csrc/hipstore.hip``):
 
 * the side that owns memory exports ``hipIpcMemHandle_t`` for the tensor's
-  caching-allocator *block* (base resolved via ``hipMemGetAddressRange``;
-  the descriptor carries the intra-block offset);
-* the peer opens the handle **once per block** (handle-bytes-keyed cache —
-  same design as the reference's weakref RdmaMemory cache,
+  caching-allocator *block* (base via ``hipMemGetAddressRange``; the
+  descriptor carries the intra-block offset);
+* the peer opens the handle **once per block** (handle-keyed cache — same
+  design as the reference's weakref RdmaMemory cache,
   ``torchcomms/cache.py:150-187``) and issues ``hipMemcpyPeerAsync`` /
   DtoD async copies on a pool of dedicated HIP streams, striping
   independent transfers across streams so multi-peer traffic aggregates
@@ -19,12 +19,23 @@ Mechanics (native side in ``csrc/ipc.cpp``):
   *push* into client-exported destination memory — both one-sided, the RPC
   only carries descriptors.
 
+**≥2 GiB blocks.** ``hipIpcOpenMemHandle`` of a ≥2³¹-byte dmabuf hangs on
+this platform (measured — the export succeeds, the peer's import never
+returns).  Tensors living in such blocks take a *windowed* path instead:
+the volume owns a pool of <2 GiB staging chunks (exported once); the client
+opens the staging and copies windows in/out of it over xGMI, with a
+handshake RPC per window committing each one.  Staging chunks assigned to
+an operation are returned to the pool when the operation's final RPC lands
+(a client that dies mid-transfer leaks its chunk until ``reset``).
+
 CPU tensors and objects in a batch ride inline in the RPC frame (an IPC
 batch can be mixed — e.g. a state_dict with scalar stats).
 """
 
 from __future__ import annotations
 
+import os
+import uuid
 from dataclasses import dataclass
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
@@ -39,6 +50,10 @@ from torchstore_amd.types import Request
 from torchstore_amd.utils.logging import get_logger
 
 logger = get_logger("torchstore_amd.hip_ipc")
+
+# blocks at or above this cannot be opened by a peer (dmabuf import hang)
+IPC_BLOCK_LIMIT = 1 << 31
+CHUNK_BYTES = int(os.environ.get("TORCHSTORE_AMD_IPC_CHUNK_MB", "512")) << 20
 
 
 @dataclass(frozen=True)
@@ -57,9 +72,29 @@ def _ext():
     return gpu.ext()
 
 
+class BlockTooLargeError(RuntimeError):
+    pass
+
+
 def export_tensor(t: torch.Tensor) -> IpcDescriptor:
+    """Export; raises :class:`BlockTooLargeError` for ≥2 GiB blocks."""
+    desc = try_export(t)
+    if desc is None:
+        raise BlockTooLargeError(
+            f"tensor lives in a >=2GiB allocator block "
+            f"({t.numel() * t.element_size()} bytes); peers cannot map it — "
+            "use the chunked transport path"
+        )
+    return desc
+
+
+def try_export(t: torch.Tensor) -> Optional[IpcDescriptor]:
     assert t.is_contiguous() and t.device.type == "cuda"
-    handle, offset = _ext().ipc_export(t.data_ptr(), t.device.index)
+    handle, offset, block_size = _ext().ipc_export(
+        t.data_ptr(), t.device.index
+    )
+    if block_size >= IPC_BLOCK_LIMIT:
+        return None
     return IpcDescriptor(
         handle=bytes(handle),
         offset=offset,
@@ -96,6 +131,45 @@ class IpcOpenCache(TransportCache):
         self.opened.clear()
 
 
+class ChunkStagingCache(TransportCache):
+    """Volume-side pool of <2 GiB staging chunks for the windowed path."""
+
+    def __init__(self):
+        self.free: List[Tuple[torch.Tensor, IpcDescriptor]] = []
+        # token -> (staging tensor, staging desc, payload tensor)
+        self.by_token: Dict[str, Tuple[torch.Tensor, IpcDescriptor, torch.Tensor]] = {}
+
+    def acquire(self, token: str, payload: torch.Tensor, device: torch.device):
+        if self.free:
+            staging, desc = self.free.pop()
+        else:
+            staging = torch.empty(CHUNK_BYTES, dtype=torch.uint8, device=device)
+            desc = export_tensor(staging)
+        self.by_token[token] = (staging, desc, payload)
+        return desc
+
+    def payload(self, token: str) -> torch.Tensor:
+        return self.by_token[token][2]
+
+    def staging(self, token: str) -> torch.Tensor:
+        return self.by_token[token][0]
+
+    def release(self, token: str) -> Optional[torch.Tensor]:
+        entry = self.by_token.pop(token, None)
+        if entry is None:
+            return None
+        staging, desc, payload = entry
+        self.free.append((staging, desc))
+        return payload
+
+    def drop_key(self, key: str) -> None:
+        return None
+
+    def close(self) -> None:
+        self.free.clear()
+        self.by_token.clear()
+
+
 def _run_copies(copies: List[Tuple[int, int, int, int, int]]) -> None:
     """(dst_ptr, dst_dev, src_ptr, src_dev, nbytes) batch on the stream pool."""
     if copies:
@@ -108,7 +182,9 @@ class HipIpcTransportBuffer(TransportBuffer):
 
     def __init__(self):
         super().__init__()
-        # aligned with requests: ("ipc", IpcDescriptor) | ("inline", value)
+        # aligned with requests:
+        #   ("ipc", IpcDescriptor) | ("inline", value) | ("chunked", token)
+        #   get-side markers: ("fetch_obj"|"fetch_inline", None)
         self.payload: Optional[List[Tuple[str, Any]]] = None
         self._hold: List[torch.Tensor] = []       # keep exports alive
         self._scratch: Dict[int, torch.Tensor] = {}  # req idx -> dense scratch
@@ -121,6 +197,95 @@ class HipIpcTransportBuffer(TransportBuffer):
         state["_hold"] = []
         state["_scratch"] = {}
         return state
+
+    # -- chunked windows (client side) -----------------------------------
+    async def _chunked_put_windows(self, t: torch.Tensor) -> str:
+        """Stream a big tensor into volume staging, window by window."""
+        volume = self._volume_ref.volume
+        cache: IpcOpenCache = self._client_ctx.cache(IpcOpenCache)
+        token = uuid.uuid4().hex
+        staging_desc = await volume.handshake.call_one(
+            self, (token, tuple(t.shape), t.dtype), "chunk_put_init"
+        )
+        staging_ptr = cache.resolve(staging_desc, t.device.index)
+        nbytes = t.numel() * t.element_size()
+        base = t.data_ptr()
+        off = 0
+        while off < nbytes:
+            win = min(CHUNK_BYTES, nbytes - off)
+            _run_copies(
+                [(staging_ptr, staging_desc.device_index,
+                  base + off, t.device.index, win)]
+            )
+            await volume.handshake.call_one(
+                self, (token, off, win), "chunk_put_commit"
+            )
+            off += win
+        return token
+
+    async def _chunked_get_windows(
+        self, request: Request, dest: torch.Tensor
+    ) -> str:
+        volume = self._volume_ref.volume
+        cache: IpcOpenCache = self._client_ctx.cache(IpcOpenCache)
+        token = uuid.uuid4().hex
+        staging_desc = await volume.handshake.call_one(
+            self, (token, request.meta_only()), "chunk_get_init"
+        )
+        staging_ptr = cache.resolve(staging_desc, dest.device.index)
+        nbytes = dest.numel() * dest.element_size()
+        base = dest.data_ptr()
+        off = 0
+        while off < nbytes:
+            win = min(CHUNK_BYTES, nbytes - off)
+            await volume.handshake.call_one(
+                self, (token, off, win), "chunk_get_fill"
+            )
+            _run_copies(
+                [(base + off, dest.device.index,
+                  staging_ptr, staging_desc.device_index, win)]
+            )
+            off += win
+        return token
+
+    # -- volume handshake dispatcher --------------------------------------
+    def recv_handshake(self, args, phase: str, volume):
+        cache: ChunkStagingCache = self._volume_ctx.cache(ChunkStagingCache)
+        device = volume.device
+        if phase == "chunk_put_init":
+            token, shape, dtype = args
+            payload = torch.empty(shape, dtype=dtype, device=device)
+            return cache.acquire(token, payload, device)
+        if phase == "chunk_put_commit":
+            token, dst_off, length = args
+            staging = cache.staging(token)
+            payload = cache.payload(token)
+            _run_copies(
+                [(payload.data_ptr() + dst_off, device.index,
+                  staging.data_ptr(), device.index, length)]
+            )
+            return "ok"
+        if phase == "chunk_get_init":
+            token, request = args
+            from torchstore_amd.ops import gpu as gpu_ops
+
+            value = volume.store.fetch(request)
+            packed = gpu_ops.pack_region(value)
+            torch.cuda.current_stream(device).synchronize()
+            return cache.acquire(token, packed, device)
+        if phase == "chunk_get_fill":
+            token, src_off, length = args
+            staging = cache.staging(token)
+            payload = cache.payload(token)
+            _run_copies(
+                [(staging.data_ptr(), device.index,
+                  payload.data_ptr() + src_off, device.index, length)]
+            )
+            return "ok"
+        if phase == "chunk_release":
+            cache.release(args)
+            return "ok"
+        raise ValueError(f"unknown handshake phase {phase!r}")
 
     # ------------------------------------------------------------- put --
     async def client_stage_put(self, requests: Sequence[Request]) -> None:
@@ -140,11 +305,17 @@ class HipIpcTransportBuffer(TransportBuffer):
                 # writes producing t must be visible before the volume pulls
                 torch.cuda.current_stream(t.device).synchronize()
                 synced.add(t.device.index)
-            payload.append(("ipc", export_tensor(tc)))
+            desc = try_export(tc)
+            if desc is None:
+                token = await self._chunked_put_windows(tc)
+                payload.append(("chunked", token))
+            else:
+                payload.append(("ipc", desc))
         self.payload = payload
 
     async def volume_receive(self, requests, existing, device):
         cache: IpcOpenCache = self._volume_ctx.cache(IpcOpenCache)
+        chunks: ChunkStagingCache = self._volume_ctx.cache(ChunkStagingCache)
         out: List[Any] = []
         copies: List[Tuple[int, int, int, int, int]] = []
         for (kind, value), prior in zip(self.payload, existing):
@@ -153,6 +324,12 @@ class HipIpcTransportBuffer(TransportBuffer):
                     out.append(value.to(device))
                 else:
                     out.append(value)
+                continue
+            if kind == "chunked":
+                payload = chunks.release(value)
+                if payload is None:
+                    raise RuntimeError("chunked put token unknown")
+                out.append(payload)
                 continue
             desc: IpcDescriptor = value
             src_ptr = cache.resolve(desc, device.index)
@@ -167,7 +344,8 @@ class HipIpcTransportBuffer(TransportBuffer):
             else:
                 dst = torch.empty(desc.shape, dtype=desc.dtype, device=device)
             copies.append(
-                (dst.data_ptr(), device.index, src_ptr, desc.device_index, desc.nbytes)
+                (dst.data_ptr(), device.index, src_ptr, desc.device_index,
+                 desc.nbytes)
             )
             out.append(dst)
         _run_copies(copies)
@@ -200,11 +378,17 @@ class HipIpcTransportBuffer(TransportBuffer):
                 torch.cuda.current_stream(target.device).synchronize()
                 synced.add(target.device.index)
             self._hold.append(target)
-            payload.append(("ipc", export_tensor(target)))
+            desc = try_export(target)
+            if desc is None:
+                token = await self._chunked_get_windows(r, target)
+                payload.append(("chunked", token))
+            else:
+                payload.append(("ipc", desc))
         self.payload = payload
 
     async def volume_send(self, requests, values):
         cache: IpcOpenCache = self._volume_ctx.cache(IpcOpenCache)
+        chunks: ChunkStagingCache = self._volume_ctx.cache(ChunkStagingCache)
         reply: List[Tuple[str, Any]] = []
         copies: List[Tuple[int, int, int, int, int]] = []
         device = None
@@ -214,6 +398,11 @@ class HipIpcTransportBuffer(TransportBuffer):
                 continue
             if kind == "fetch_inline" or v.device.type != "cuda":
                 reply.append(("inline", v))
+                continue
+            if kind == "chunked":
+                # windows already delivered during the handshake phases
+                chunks.release(value)
+                reply.append(("done", None))
                 continue
             desc: IpcDescriptor = value
             device = v.device
