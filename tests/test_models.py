@@ -71,7 +71,7 @@ async def test_llama8b_shapes_layer_subset():
 
     async def body():
         sd = llama.make_sharded_state_dict(
-            None, llama.fsdp_placement, device="cpu", layers=1, seed=3
+            None, llama.fsdp_placement, device="cpu", layers=1, seed=3, scale=8
         )
         await ts.put_state_dict(sd, "m8b")
         out = await ts.get_state_dict("m8b")

@@ -19,23 +19,27 @@ VOCAB = 128256
 LAYERS = 32
 
 
-def llama3_8b_shapes(layers: int = LAYERS) -> Dict[str, Tuple[int, ...]]:
+def llama3_8b_shapes(
+    layers: int = LAYERS, scale: int = 1
+) -> Dict[str, Tuple[int, ...]]:
+    """Exact Llama-3-8B shapes; ``scale`` > 1 shrinks every dim (tests)."""
+    v, h, kv, it = VOCAB // scale, HIDDEN // scale, KV // scale, INTER // scale
     shapes: Dict[str, Tuple[int, ...]] = {
-        "model.embed_tokens.weight": (VOCAB, HIDDEN),
+        "model.embed_tokens.weight": (v, h),
     }
     for i in range(layers):
         p = f"model.layers.{i}."
-        shapes[p + "self_attn.q_proj.weight"] = (HIDDEN, HIDDEN)
-        shapes[p + "self_attn.k_proj.weight"] = (KV, HIDDEN)
-        shapes[p + "self_attn.v_proj.weight"] = (KV, HIDDEN)
-        shapes[p + "self_attn.o_proj.weight"] = (HIDDEN, HIDDEN)
-        shapes[p + "mlp.gate_proj.weight"] = (INTER, HIDDEN)
-        shapes[p + "mlp.up_proj.weight"] = (INTER, HIDDEN)
-        shapes[p + "mlp.down_proj.weight"] = (HIDDEN, INTER)
-        shapes[p + "input_layernorm.weight"] = (HIDDEN,)
-        shapes[p + "post_attention_layernorm.weight"] = (HIDDEN,)
-    shapes["model.norm.weight"] = (HIDDEN,)
-    shapes["lm_head.weight"] = (VOCAB, HIDDEN)
+        shapes[p + "self_attn.q_proj.weight"] = (h, h)
+        shapes[p + "self_attn.k_proj.weight"] = (kv, h)
+        shapes[p + "self_attn.v_proj.weight"] = (kv, h)
+        shapes[p + "self_attn.o_proj.weight"] = (h, h)
+        shapes[p + "mlp.gate_proj.weight"] = (it, h)
+        shapes[p + "mlp.up_proj.weight"] = (it, h)
+        shapes[p + "mlp.down_proj.weight"] = (h, it)
+        shapes[p + "input_layernorm.weight"] = (h,)
+        shapes[p + "post_attention_layernorm.weight"] = (h,)
+    shapes["model.norm.weight"] = (h,)
+    shapes["lm_head.weight"] = (v, h)
     return shapes
 
 
@@ -73,6 +77,7 @@ def make_local_shard_state_dict(
     device: str = "cuda",
     layers: int = LAYERS,
     zero: bool = True,
+    scale: int = 1,
 ):
     """DTensor-free sharded state_dict: {name: LocalShard | tensor}.
 
@@ -82,7 +87,7 @@ def make_local_shard_state_dict(
     """
     from torchstore_amd.types import LocalShard, TensorSlice
 
-    shapes = llama3_8b_shapes(layers)
+    shapes = llama3_8b_shapes(layers, scale)
     out = {}
     for name, shape in shapes.items():
         if world == 1:
@@ -125,6 +130,7 @@ def make_sharded_state_dict(
     layers: int = LAYERS,
     zero: bool = False,
     seed: Optional[int] = None,
+    scale: int = 1,
 ) -> Dict[str, torch.Tensor]:
     """Build {name: DTensor|tensor} where each rank materializes only its shard.
 
@@ -136,7 +142,7 @@ def make_sharded_state_dict(
         compute_local_shape_and_global_offset,
     )
 
-    shapes = llama3_8b_shapes(layers)
+    shapes = llama3_8b_shapes(layers, scale)
     out: Dict[str, torch.Tensor] = {}
     world = mesh.size() if mesh is not None else 1
     gen = None
