@@ -167,34 +167,16 @@ static void ipc_close(uintptr_t base, int local_device) {
 // ---------------------------------------------------------------------------
 
 // copies: (dst_ptr, dst_dev, src_ptr, src_dev, nbytes)
+//
+// Same-device copies batch into ONE copy_slices kernel launch per device
+// (291 sequential hipMemcpyAsync enqueues cost ~20 us each on the host —
+// the kernel path replaces them with one dispatch).  Cross-device copies
+// keep hipMemcpyPeerAsync striped round-robin over the stream pool so
+// transfers to different peers ride different xGMI links concurrently.
+// Definition follows the slice kernel below.
 static void copy_batch(
     const std::vector<std::tuple<uintptr_t, int, uintptr_t, int, uint64_t>>&
-        copies) {
-  if (copies.empty()) return;
-  std::vector<hipStream_t> used;
-  int i = 0;
-  for (const auto& c : copies) {
-    uintptr_t dst = std::get<0>(c);
-    int dst_dev = std::get<1>(c);
-    uintptr_t src = std::get<2>(c);
-    int src_dev = std::get<3>(c);
-    uint64_t n = std::get<4>(c);
-    DevicePool& p = pool_for(dst_dev);
-    hipStream_t s = p.streams[i++ % kStreamsPerDevice];
-    HIP_CHECK(hipSetDevice(dst_dev));
-    if (dst_dev == src_dev) {
-      HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(dst),
-                               reinterpret_cast<void*>(src), n,
-                               hipMemcpyDeviceToDevice, s));
-    } else {
-      HIP_CHECK(hipMemcpyPeerAsync(reinterpret_cast<void*>(dst), dst_dev,
-                                   reinterpret_cast<void*>(src), src_dev, n,
-                                   s));
-    }
-    used.push_back(s);
-  }
-  for (hipStream_t s : used) HIP_CHECK(hipStreamSynchronize(s));
-}
+        copies);
 
 // copies: (dst_ptr, dst_dev, dpitch, src_ptr, src_dev, spitch, width, height)
 // — strided (pitched) one-sided reads/writes: moves ONLY the overlap bytes
@@ -326,6 +308,89 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
   }
 }
 
+// upload descriptors through the pinned staging buffer + launch the kernel
+static void launch_slice_descs(std::vector<SliceDesc>& descs, uint64_t units,
+                               int device, hipStream_t stream) {
+  if (units == 0 || descs.empty()) return;
+  DevicePool& p = pool_for(device);
+  HIP_CHECK(hipSetDevice(device));
+  size_t bytes = descs.size() * sizeof(SliceDesc);
+  std::lock_guard<std::mutex> lock(g_mutex);
+  ensure_desc_capacity(p, device, bytes);
+  if (p.desc_evt == nullptr) {
+    HIP_CHECK(hipEventCreateWithFlags(&p.desc_evt, hipEventDisableTiming));
+  } else {
+    // the previous call's KERNEL must be done before h_desc/d_desc are
+    // reused (calls may target different streams)
+    HIP_CHECK(hipEventSynchronize(p.desc_evt));
+  }
+  std::memcpy(p.h_desc, descs.data(), bytes);
+  HIP_CHECK(hipMemcpyAsync(p.d_desc, p.h_desc, bytes, hipMemcpyHostToDevice,
+                           stream));
+  // memory-bound: cap grid at 2048 blocks, grid-stride the rest (guide
+  // G11); each block consumes 4 units (one per wave)
+  uint32_t grid = (uint32_t)std::min<uint64_t>((units + 3) / 4, 2048);
+  hipLaunchKernelGGL(copy_slices_kernel, dim3(grid), dim3(256), 0, stream,
+                     reinterpret_cast<const SliceDesc*>(p.d_desc),
+                     (uint32_t)descs.size(), units);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipEventRecord(p.desc_evt, stream));
+}
+
+// trivial desc for a flat byte copy (used by copy_batch's same-device path)
+static SliceDesc flat_desc(uintptr_t dst, uintptr_t src, uint64_t nbytes,
+                           uint64_t units_prefix) {
+  SliceDesc d{};
+  d.src = src;
+  d.dst = dst;
+  d.rows = 1;
+  d.units_prefix = units_prefix;
+  d.row_bytes = (uint32_t)nbytes;
+  d.tiles_per_row = (uint32_t)((nbytes + kTileBytes - 1) / kTileBytes);
+  if (d.tiles_per_row == 0) d.tiles_per_row = 1;
+  d.ndim = 0;
+  return d;
+}
+
+static void copy_batch(
+    const std::vector<std::tuple<uintptr_t, int, uintptr_t, int, uint64_t>>&
+        copies) {
+  if (copies.empty()) return;
+  // group same-device copies per device; cross-device go via SDMA
+  std::unordered_map<int, std::vector<SliceDesc>> per_device;
+  std::unordered_map<int, uint64_t> per_device_units;
+  std::vector<hipStream_t> used;
+  int i = 0;
+  for (const auto& c : copies) {
+    uintptr_t dst = std::get<0>(c);
+    int dst_dev = std::get<1>(c);
+    uintptr_t src = std::get<2>(c);
+    int src_dev = std::get<3>(c);
+    uint64_t n = std::get<4>(c);
+    if (n == 0) continue;
+    if (dst_dev == src_dev && n <= UINT32_MAX) {
+      uint64_t& units = per_device_units[dst_dev];
+      per_device[dst_dev].push_back(flat_desc(dst, src, n, units));
+      units += per_device[dst_dev].back().tiles_per_row;
+      continue;
+    }
+    DevicePool& p = pool_for(dst_dev);
+    hipStream_t s = p.streams[i++ % kStreamsPerDevice];
+    HIP_CHECK(hipSetDevice(dst_dev));
+    HIP_CHECK(hipMemcpyPeerAsync(reinterpret_cast<void*>(dst), dst_dev,
+                                 reinterpret_cast<void*>(src), src_dev, n, s));
+    used.push_back(s);
+  }
+  for (auto& kv : per_device) {
+    int device = kv.first;
+    DevicePool& p = pool_for(device);
+    hipStream_t s = p.streams[0];
+    launch_slice_descs(kv.second, per_device_units[device], device, s);
+    used.push_back(s);
+  }
+  for (hipStream_t s : used) HIP_CHECK(hipStreamSynchronize(s));
+}
+
 // python passes per slice:
 //   (src_ptr, dst_ptr, row_bytes, [outer shape], [src strides B], [dst strides B])
 using PySlice = std::tuple<uintptr_t, uintptr_t, uint64_t,
@@ -365,33 +430,12 @@ static void copy_slices(const std::vector<PySlice>& slices, int device,
   }
   if (units == 0) return;
 
-  DevicePool& p = pool_for(device);
-  HIP_CHECK(hipSetDevice(device));
-  size_t bytes = n * sizeof(SliceDesc);
-  hipStream_t stream = reinterpret_cast<hipStream_t>(stream_handle);
-  uint32_t grid = 0;
-  {
-    std::lock_guard<std::mutex> lock(g_mutex);
-    ensure_desc_capacity(p, device, bytes);
-    if (p.desc_evt == nullptr) {
-      HIP_CHECK(hipEventCreateWithFlags(&p.desc_evt, hipEventDisableTiming));
-    } else {
-      // previous call's staging H2D must be done before h_desc is reused
-      HIP_CHECK(hipEventSynchronize(p.desc_evt));
-    }
-    std::memcpy(p.h_desc, descs.data(), bytes);
-    HIP_CHECK(hipMemcpyAsync(p.d_desc, p.h_desc, bytes, hipMemcpyHostToDevice,
-                             stream));
-    HIP_CHECK(hipEventRecord(p.desc_evt, stream));
-    // memory-bound: cap grid at 2048 blocks, grid-stride the rest (guide
-    // G11); each block consumes 4 units (one per wave)
-    grid = (uint32_t)std::min<uint64_t>((units + 3) / 4, 2048);
-    hipLaunchKernelGGL(copy_slices_kernel, dim3(grid), dim3(256), 0, stream,
-                       reinterpret_cast<const SliceDesc*>(p.d_desc),
-                       (uint32_t)n, units);
-    HIP_CHECK(hipGetLastError());
+  launch_slice_descs(descs, units, device,
+                     reinterpret_cast<hipStream_t>(stream_handle));
+  if (blocking) {
+    HIP_CHECK(
+        hipStreamSynchronize(reinterpret_cast<hipStream_t>(stream_handle)));
   }
-  if (blocking) HIP_CHECK(hipStreamSynchronize(stream));
 }
 
 // ---------------------------------------------------------------------------
