@@ -1,0 +1,77 @@
+"""Failure semantics: dead volumes, idempotent deletes, tolerant exists,
+reset, and teardown robustness (SURVEY §5.3 — the reference's level of
+fault handling plus cleaner surfacing)."""
+
+import asyncio
+
+import pytest
+import torch
+
+import torchstore_amd as ts
+from torchstore_amd.strategy import SingletonStrategy
+from torchstore_amd.transport import TransportType
+
+
+async def test_dead_volume_surfaces_connection_error():
+    await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(transport=TransportType.RPC),
+        storage_device="cpu",
+    )
+    try:
+        await ts.put("k", torch.ones(4))
+        # kill the volume process out from under the store
+        session = ts.api._sessions["default"]
+        for p in session.volume_mesh._procs:
+            p.terminate()
+            p.join(timeout=10)
+        with pytest.raises((ConnectionError, OSError)):
+            await ts.get("k")
+    finally:
+        await ts.shutdown()
+
+
+async def test_exists_never_raises_for_missing():
+    await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(),
+        storage_device="cpu",
+    )
+    try:
+        assert not await ts.exists("missing/deeply/nested")
+        assert await ts.keys("missing") == []
+    finally:
+        await ts.shutdown()
+
+
+async def test_reset_clears_everything():
+    await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(),
+        storage_device="cpu",
+    )
+    try:
+        await ts.put("a", torch.ones(4))
+        session = ts.api._sessions["default"]
+        await session.controller.teardown.call_one()
+        assert await ts.keys() == []
+        with pytest.raises(KeyError):
+            await ts.get("a")
+        # the store is usable again after teardown
+        await ts.put("b", torch.ones(2))
+        assert (await ts.get("b")).eq(1).all()
+    finally:
+        await ts.shutdown()
+
+
+async def test_double_shutdown_is_safe():
+    await ts.initialize(
+        num_storage_volumes=1, strategy=SingletonStrategy(), storage_device="cpu"
+    )
+    await ts.shutdown()
+    await ts.shutdown()  # second time: no-op
+
+
+async def test_client_before_initialize_raises():
+    with pytest.raises(RuntimeError, match="not initialized"):
+        ts.client("never-made")

@@ -101,13 +101,16 @@ class LocalClient:
         volume_id = self._strategy.select_volume_id(list(volumes.keys()))
         ref = self._volume_ref(volume_id)
         tracker = LatencyTracker(f"put_batch[{len(requests)}]")
-        buffer = create_transport(ref)
-        await buffer.put(requests)
-        tracker.step("transport", sum(r.nbytes() for r in requests))
-        await self._controller.notify_put_batch.call_one(
-            [r.meta_only() for r in requests], volume_id
-        )
-        tracker.step("notify")
+        from torchstore_amd.utils.logging import roctx_range
+
+        with roctx_range(f"ts::put_batch[{len(requests)}]"):
+            buffer = create_transport(ref)
+            await buffer.put(requests)
+            tracker.step("transport", sum(r.nbytes() for r in requests))
+            await self._controller.notify_put_batch.call_one(
+                [r.meta_only() for r in requests], volume_id
+            )
+            tracker.step("notify")
 
     # -- get --------------------------------------------------------------
     async def get(self, key: str, like: Any = None) -> Any:
@@ -263,9 +266,12 @@ class LocalClient:
         return torch.device("cpu")
 
     async def _fetch_volume(self, volume_id: str, subs: List[_SubFetch]) -> None:
+        from torchstore_amd.utils.logging import roctx_range
+
         ref = self._volume_ref(volume_id)
-        buffer = create_transport(ref)
-        results = await buffer.get([sf.request for sf in subs])
+        with roctx_range(f"ts::get[v{volume_id}:{len(subs)}]"):
+            buffer = create_transport(ref)
+            results = await buffer.get([sf.request for sf in subs])
         for sf, res in zip(subs, results):
             sf.result = res
 
