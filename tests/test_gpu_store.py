@@ -146,6 +146,54 @@ async def test_gpu_state_dict_roundtrip():
 
 
 @requires_gpu
+async def test_two_volumes_one_gpu_fanout():
+    """Two GPU-resident volumes on one device: shard puts route to their
+    own volume; a full get fans out to both concurrently and assembles —
+    the single-box stand-in for the 8-volume xGMI topology."""
+    import os
+
+    from torchstore_amd.strategy import LocalRankStrategy
+    from torchstore_amd.types import LocalShard, TensorSlice
+
+    await ts.initialize(
+        num_storage_volumes=2,
+        strategy=LocalRankStrategy(transport=TransportType.HIP_IPC),
+        storage_device="auto",
+    )
+    try:
+        full = torch.randn(512, 512, device="cuda")
+        for rank in range(2):
+            os.environ["RANK"] = str(rank)  # place each shard on its volume
+            shard = LocalShard(
+                tensor=full[rank * 256 : (rank + 1) * 256].contiguous(),
+                slice=TensorSlice(
+                    offsets=(rank * 256, 0), local_shape=(256, 512),
+                    global_shape=(512, 512), coordinates=(rank,),
+                    mesh_shape=(2,),
+                ),
+            )
+            await ts.put("w", shard)
+        out = await ts.get("w")
+        torch.cuda.synchronize()
+        assert out.device.type == "cuda"
+        assert torch.equal(out, full)
+        # column-slice dest: both volumes write strided pieces via scratch+K2
+        dest = LocalShard(
+            tensor=torch.zeros(512, 256, device="cuda"),
+            slice=TensorSlice(
+                offsets=(0, 128), local_shape=(512, 256),
+                global_shape=(512, 512), coordinates=(0,), mesh_shape=(2,),
+            ),
+        )
+        await ts.get("w", dest)
+        torch.cuda.synchronize()
+        assert torch.equal(dest.tensor, full[:, 128:384])
+    finally:
+        os.environ.pop("RANK", None)
+        await ts.shutdown()
+
+
+@requires_gpu
 async def test_gpu_put_does_not_sync_foreign_stream():
     """The reference's stream-isolation invariant: a put must not wait on
     unrelated work queued on another stream (test_shared_memory.py:1034)."""

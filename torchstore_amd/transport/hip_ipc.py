@@ -348,7 +348,12 @@ class HipIpcTransportBuffer(TransportBuffer):
                  desc.nbytes)
             )
             out.append(dst)
-        _run_copies(copies)
+        if copies:
+            # executor thread: the volume keeps serving other clients while
+            # the batched pull runs (the C++ side drops the GIL)
+            import asyncio
+
+            await asyncio.to_thread(_run_copies, copies)
         return out
 
     # ------------------------------------------------------------- get --
@@ -425,7 +430,10 @@ class HipIpcTransportBuffer(TransportBuffer):
             # K1 pack kernels ran on the current stream; the pool streams
             # used by copy_batch must observe their writes
             torch.cuda.current_stream(device).synchronize()
-        _run_copies(copies)
+        if copies:
+            import asyncio
+
+            await asyncio.to_thread(_run_copies, copies)
         return reply
 
     def client_complete_get(self, requests, reply) -> List[Any]:
