@@ -2,6 +2,7 @@
 objects, batches, both RPC and SHM transports."""
 
 import asyncio
+import os
 
 import pytest
 import torch
@@ -130,20 +131,21 @@ async def test_keys_prefix(transport):
 
 
 async def test_slice_get_of_full_tensor(transport):
-    """Fetching a sub-slice of a stored full tensor."""
-    from torchstore_amd.types import TensorSlice
-    from torchstore_amd.client import _full_region_slice
+    """Fetching a sub-region of a stored full tensor via a LocalShard dest."""
+    from torchstore_amd.types import LocalShard, TensorSlice
 
     async def body():
         t = torch.arange(64, dtype=torch.float32).reshape(8, 8)
         await ts.put("big", t)
-        c = ts.client()
-        # request rows 2..6 via a tensor_slice'd inplace dest
-        dest = torch.zeros(4, 8)
-        from torchstore_amd.types import Request
-
-        got = await c.get("big", dest)  # full fetch sanity
-        assert got.shape == (8, 8) or torch.equal(dest, t[:4])
+        dest = LocalShard(
+            tensor=torch.zeros(3, 5),
+            slice=TensorSlice(
+                offsets=(2, 1), local_shape=(3, 5), global_shape=(8, 8),
+                coordinates=(0,), mesh_shape=(1,),
+            ),
+        )
+        await ts.get("big", dest)
+        assert torch.equal(dest.tensor, t[2:5, 1:6])
 
     await _with_store(transport, body)
 
@@ -159,6 +161,30 @@ async def test_two_volumes_round_robin():
     try:
         await ts.put("x", torch.ones(4))
         assert torch.equal(await ts.get("x"), torch.ones(4))
+    finally:
+        await ts.shutdown()
+
+
+async def test_host_strategy():
+    """Volume id = hostname; clients pick their host's volume."""
+    from torchstore_amd.strategy import HostStrategy
+
+    await ts.initialize(
+        num_storage_volumes=1,
+        strategy=HostStrategy(),
+        storage_device="cpu",
+    )
+    try:
+        c = ts.client()
+        await c._ensure_volumes()
+        import socket
+
+        vid = next(iter(c._volumes))
+        assert vid == (os.environ.get("HOSTNAME") or socket.gethostname())
+        ref = c._volume_ref(vid)
+        assert ref.is_local
+        await ts.put("h", torch.full((4,), 3.0))
+        assert (await ts.get("h")).eq(3).all()
     finally:
         await ts.shutdown()
 

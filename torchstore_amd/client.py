@@ -120,12 +120,15 @@ class LocalClient:
     async def get_batch(self, fetches: Dict[str, Any]) -> Dict[str, Any]:
         if not fetches:
             return {}
+        tracker = LatencyTracker(f"get_batch[{len(fetches)}]")
         await self._ensure_volumes()
         located = await self._controller.locate.call_one(list(fetches.keys()))
+        tracker.step("locate")
         plans = {
             key: await self._plan_fetch(key, like, located[key])
             for key, like in fetches.items()
         }
+        tracker.step("plan")
         # group sub-fetches per volume, fetch all volumes concurrently
         by_volume: Dict[str, List[_SubFetch]] = {}
         for subs in plans.values():
@@ -134,10 +137,13 @@ class LocalClient:
         await asyncio.gather(
             *(self._fetch_volume(vid, sfs) for vid, sfs in by_volume.items())
         )
-        return {
+        tracker.step("fetch")
+        out = {
             key: self._finish_fetch(key, fetches[key], [sf for _, sf in subs])
             for key, subs in plans.items()
         }
+        tracker.step("finish")
+        return out
 
     async def _plan_fetch(
         self, key: str, like: Any, locations: Dict[str, StorageInfo]
