@@ -137,6 +137,52 @@ def copy_pairs(
         ext().copy_slices(descs, device.index, _stream(device), blocking)
 
 
+def _desc_view_to_ptr(src: torch.Tensor, dst_ptr: int):
+    """Descriptor: strided src view → contiguous bytes at a raw pointer.
+
+    Fuses K1 (gather) with the transfer itself: the kernel reads the
+    stored view and writes straight into the (same-device) peer-mapped
+    destination — no packed intermediate, half the HBM traffic.
+    """
+    if src.numel() == 0:
+        return ()
+    es = src.element_size()
+    if src.dim() == 0:
+        return (src.data_ptr(), dst_ptr, es, [], [], [])
+    if src.stride(-1) != 1:
+        return None
+    row_bytes = src.shape[-1] * es
+    outer = list(src.shape[:-1])
+    sst = [s * es for s in src.stride()[:-1]]
+    # contiguous destination strides for src.shape
+    dstst = []
+    acc = row_bytes
+    for d in reversed(outer):
+        dstst.append(acc)
+        acc *= d
+    dstst = list(reversed(dstst))
+    return (src.data_ptr(), dst_ptr, row_bytes, outer, sst, dstst)
+
+
+def copy_views_to_ptrs(items, device: torch.device, blocking: bool = True):
+    """Batched fused gather+write: ``[(src_view, dst_ptr), ...]``.
+
+    Returns the list of items the kernel could NOT express (caller falls
+    back to pack+copy for those).
+    """
+    descs = []
+    rejects = []
+    for src, dst_ptr in items:
+        d = _desc_view_to_ptr(src, dst_ptr)
+        if d is None:
+            rejects.append((src, dst_ptr))
+        elif d != ():
+            descs.append(d)
+    if descs:
+        ext().copy_slices(descs, device.index, _stream(device), blocking)
+    return rejects
+
+
 def pack_region(src_view: torch.Tensor) -> torch.Tensor:
     """K1: strided region → freshly-allocated contiguous tensor."""
     if src_view.is_contiguous():

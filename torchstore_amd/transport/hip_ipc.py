@@ -394,8 +394,11 @@ class HipIpcTransportBuffer(TransportBuffer):
     async def volume_send(self, requests, values):
         cache: IpcOpenCache = self._volume_ctx.cache(IpcOpenCache)
         chunks: ChunkStagingCache = self._volume_ctx.cache(ChunkStagingCache)
+        from torchstore_amd.ops import gpu as gpu_ops
+
         reply: List[Tuple[str, Any]] = []
-        copies: List[Tuple[int, int, int, int, int]] = []
+        fused: List[Tuple[torch.Tensor, int]] = []   # same-device: K1 writes
+        copies: List[Tuple[int, int, int, int, int]] = []  # cross-device SDMA
         device = None
         for (kind, value), r, v in zip(self.payload, requests, values):
             if kind == "fetch_obj" or not isinstance(v, torch.Tensor):
@@ -411,26 +414,43 @@ class HipIpcTransportBuffer(TransportBuffer):
                 continue
             desc: IpcDescriptor = value
             device = v.device
-            from torchstore_amd.ops import gpu as gpu_ops
-
-            vc = gpu_ops.pack_region(v)  # K1 slice gather for strided views
-            if vc.numel() * vc.element_size() != desc.nbytes:
+            if v.numel() * v.element_size() != desc.nbytes:
                 raise RuntimeError(
-                    f"get size mismatch for {r.key}: stored {vc.shape} vs "
-                    f"dest {desc.shape}"
+                    f"get size mismatch for {r.key}: stored {tuple(v.shape)} "
+                    f"vs dest {desc.shape}"
                 )
             dst_ptr = cache.resolve(desc, v.device.index)
-            copies.append(
-                (dst_ptr, desc.device_index, vc.data_ptr(), v.device.index,
-                 desc.nbytes)
-            )
+            if desc.device_index == v.device.index:
+                # co-located: the K1 gather kernel writes STRAIGHT into the
+                # client's mapped destination — no packed intermediate
+                fused.append((v, dst_ptr))
+            else:
+                vc = gpu_ops.pack_region(v)  # K1 gather for strided views
+                copies.append(
+                    (dst_ptr, desc.device_index, vc.data_ptr(),
+                     v.device.index, desc.nbytes)
+                )
+                self._hold.append(vc)
             reply.append(("done", None))
-            self._hold.append(vc)
-        if copies and device is not None:
+        if fused:
+            import asyncio
+
+            # executor thread: loop stays responsive; kernel runs on the
+            # device's default stream and is synchronized before return
+            rejects = await asyncio.to_thread(
+                gpu_ops.copy_views_to_ptrs, fused, device, True
+            )
+            for v, dst_ptr in rejects:  # kernel-inexpressible layout
+                vc = gpu_ops.pack_region(v)
+                copies.append(
+                    (dst_ptr, device.index, vc.data_ptr(), device.index,
+                     vc.numel() * vc.element_size())
+                )
+                self._hold.append(vc)
+        if copies:
             # K1 pack kernels ran on the current stream; the pool streams
             # used by copy_batch must observe their writes
             torch.cuda.current_stream(device).synchronize()
-        if copies:
             import asyncio
 
             await asyncio.to_thread(_run_copies, copies)
