@@ -109,6 +109,12 @@ def _slice_desc(src: torch.Tensor, dst: torch.Tensor):
     es = src.element_size()
     if src.dim() == 0:
         return (src.data_ptr(), dst.data_ptr(), es, [], [], [])
+    if (
+        src.is_contiguous()
+        and dst.is_contiguous()
+        and src.numel() * es < (1 << 32)
+    ):
+        return (src.data_ptr(), dst.data_ptr(), src.numel() * es, [], [], [])
     if src.stride(-1) != 1 or dst.stride(-1) != 1:
         return None
     row_bytes = src.shape[-1] * es
@@ -147,8 +153,12 @@ def _desc_view_to_ptr(src: torch.Tensor, dst_ptr: int):
     if src.numel() == 0:
         return ()
     es = src.element_size()
+    nbytes = src.numel() * es
     if src.dim() == 0:
         return (src.data_ptr(), dst_ptr, es, [], [], [])
+    if src.is_contiguous():
+        # flat descriptor: full 16 KiB tiles, no per-row offset math
+        return (src.data_ptr(), dst_ptr, nbytes, [], [], [])
     if src.stride(-1) != 1:
         return None
     row_bytes = src.shape[-1] * es
