@@ -201,7 +201,7 @@ static void copy_batch_2d(
     HIP_CHECK(hipSetDevice(dst_dev));
     HIP_CHECK(hipMemcpy2DAsync(reinterpret_cast<void*>(dst), dpitch,
                                reinterpret_cast<void*>(src), spitch, width,
-                               height, hipMemcpyDeviceToDevice, s));
+                               height, hipMemcpyDefault, s));
     used.push_back(s);
   }
   for (hipStream_t s : used) HIP_CHECK(hipStreamSynchronize(s));
@@ -377,8 +377,13 @@ static void copy_batch(
     DevicePool& p = pool_for(dst_dev);
     hipStream_t s = p.streams[i++ % kStreamsPerDevice];
     HIP_CHECK(hipSetDevice(dst_dev));
-    HIP_CHECK(hipMemcpyPeerAsync(reinterpret_cast<void*>(dst), dst_dev,
-                                 reinterpret_cast<void*>(src), src_dev, n, s));
+    // UVA-resolved copy: the canonical form for IPC-mapped peer pointers
+    // (hipMemcpyPeerAsync requires context-owned pointers on both ends;
+    // Default kind lets the driver resolve mapped peer memory)
+    HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(dst),
+                             reinterpret_cast<void*>(src), n,
+                             hipMemcpyDefault, s));
+    (void)src_dev;
     used.push_back(s);
   }
   for (auto& kv : per_device) {

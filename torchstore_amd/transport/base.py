@@ -78,7 +78,12 @@ class TransportContext:
     """
 
     def __init__(self):
+        import uuid
+
         self._caches: Dict[type, TransportCache] = {}
+        # stable per-context identity (e.g. volume-side caches key client
+        # resources by it so concurrent clients never share buffers)
+        self.uid = uuid.uuid4().hex
 
     def cache(self, cls: Type[TransportCache]) -> TransportCache:
         inst = self._caches.get(cls)

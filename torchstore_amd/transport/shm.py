@@ -221,11 +221,9 @@ class ShmVolumeCache(TransportCache):
         return entry
 
     def drop_key(self, key: str) -> None:
-        entry = self.put_segments.pop(key, None)
-        if entry is not None:
-            self.desc_by_storage.pop(
-                entry[1].untyped_storage().data_ptr(), None
-            )
+        for k in [k for k in self.put_segments if k[1] == key]:
+            desc, seg = self.put_segments.pop(k)
+            self.desc_by_storage.pop(seg.untyped_storage().data_ptr(), None)
         for k in [k for k in self.get_segments if k[0] == key]:
             desc, seg = self.get_segments.pop(k)
             self.desc_by_storage.pop(seg.untyped_storage().data_ptr(), None)
@@ -248,6 +246,9 @@ class ShmTransportBuffer(TransportBuffer):
         self.payload: Optional[List[Tuple[str, Any]]] = None
         # put handshake: per-request nbytes to allocate (None for objects)
         self.alloc_sizes: Optional[List[Optional[int]]] = None
+        # client identity: volume-side put segments are keyed (client, key)
+        # so concurrent clients writing one key never share a segment
+        self.client_uid: str = ""
 
     # -- handshake (volume allocates put segments) ------------------------
     def recv_handshake(self, requests: Sequence[Request], phase: str, volume):
@@ -259,7 +260,9 @@ class ShmTransportBuffer(TransportBuffer):
             if nbytes is None:
                 out.append(None)
                 continue
-            desc, _seg = cache.obtain(cache.put_segments, r.key, nbytes)
+            desc, _seg = cache.obtain(
+                cache.put_segments, (self.client_uid, r.key), nbytes
+            )
             out.append(desc)
         return out
 
@@ -270,6 +273,7 @@ class ShmTransportBuffer(TransportBuffer):
         self.alloc_sizes = [
             None if r.is_object else r.nbytes() for r in requests
         ]
+        self.client_uid = self._client_ctx.uid
         try:
             reply = await volume.handshake.call_one(self, metas, "put")
             cache: ShmClientCache = self._client_ctx.cache(ShmClientCache)
@@ -306,7 +310,7 @@ class ShmTransportBuffer(TransportBuffer):
                 out.append(value)
                 continue
             desc: ShmDescriptor = value
-            cached = cache.put_segments.get(r.key)
+            cached = cache.put_segments.get((self.client_uid, r.key))
             if cached is not None and cached[0].seg_key == desc.seg_key:
                 seg = cached[1]
             else:
