@@ -242,7 +242,17 @@ def main():
     if not torch.cuda.is_available():
         raise SystemExit("bench.py needs MI355X GPUs (torch.cuda unavailable)")
     rank, world, local_rank = setup_dist(args)
-    asyncio.run(run_bench(args, rank, world, local_rank))
+    try:
+        asyncio.run(run_bench(args, rank, world, local_rank))
+    except Exception:
+        import sys
+        import traceback
+
+        print(
+            f"[bench rank {rank}/{world}] FAILED:\n{traceback.format_exc()}",
+            file=sys.stderr, flush=True,
+        )
+        raise
 
 
 if __name__ == "__main__":

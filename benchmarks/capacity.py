@@ -1,0 +1,81 @@
+"""Capacity benchmark: many-GB put_batch/get_batch across GPU-resident
+volumes (the BASELINE '200 GB across 8 volumes' config, scaled to the
+available GPUs — 288 GB HBM per MI355X holds client + volume copies).
+
+    python benchmarks/capacity.py --total-gb 100 --chunk-mb 1024
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+import torchstore_amd as ts
+from torchstore_amd.strategy import LocalRankStrategy
+
+
+async def run(total_gb: float, chunk_mb: int, verify: bool):
+    await ts.initialize(
+        num_storage_volumes=1,
+        strategy=LocalRankStrategy(),
+        storage_device="auto",
+    )
+    try:
+        chunk_bytes = chunk_mb << 20
+        n = max(1, int(total_gb * 1e9 / chunk_bytes))
+        numel = chunk_bytes // 2
+        print(f"payload: {n} x {chunk_mb} MB bf16 chunks "
+              f"({n * chunk_bytes / 1e9:.1f} GB)", flush=True)
+        items = {}
+        for i in range(n):
+            t = torch.empty(numel, dtype=torch.bfloat16, device="cuda")
+            t.view(torch.int16).fill_(i % 251)
+            items[f"cap/{i}"] = t
+        torch.cuda.synchronize()
+
+        t0 = time.perf_counter()
+        await ts.put_batch(items)
+        torch.cuda.synchronize()
+        put_dt = time.perf_counter() - t0
+        total = n * chunk_bytes
+        print(f"put_batch: {put_dt*1e3:.1f} ms  {total/put_dt/1e9:.0f} GB/s",
+              flush=True)
+        free, cap = torch.cuda.mem_get_info()
+        print(f"client HBM used: {(cap - free)/1e9:.1f} / {cap/1e9:.0f} GB",
+              flush=True)
+
+        dests = {k: torch.empty_like(v) for k, v in items.items()}
+        t0 = time.perf_counter()
+        out = await ts.get_batch(dests)
+        torch.cuda.synchronize()
+        get_dt = time.perf_counter() - t0
+        print(f"get_batch: {get_dt*1e3:.1f} ms  {total/get_dt/1e9:.0f} GB/s",
+              flush=True)
+        if verify:
+            for i in (0, n // 2, n - 1):
+                k = f"cap/{i}"
+                assert torch.equal(out[k], items[k]), k
+            print("verify ok", flush=True)
+    finally:
+        await ts.shutdown()
+
+
+def main():
+    p = argparse.ArgumentParser()
+    # src + stored + dest copies coexist on one 288 GB GPU
+    p.add_argument("--total-gb", type=float, default=80.0)
+    p.add_argument("--chunk-mb", type=int, default=1024)
+    p.add_argument("--no-verify", action="store_true")
+    args = p.parse_args()
+    asyncio.run(run(args.total_gb, args.chunk_mb, not args.no_verify))
+
+
+if __name__ == "__main__":
+    main()

@@ -281,6 +281,26 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
       uint4* d4 = reinterpret_cast<uint4*>(dst);
       uint32_t n4 = len >> 4;
       uint32_t i = lane;
+      // x8 unroll: 128 B of loads in flight per lane before the first
+      // dependent store (HBM latency ~300 cyc wants deep MLP)
+      for (; i + 448 < n4; i += 512) {
+        uint4 v0 = s4[i];
+        uint4 v1 = s4[i + 64];
+        uint4 v2 = s4[i + 128];
+        uint4 v3 = s4[i + 192];
+        uint4 v4 = s4[i + 256];
+        uint4 v5 = s4[i + 320];
+        uint4 v6 = s4[i + 384];
+        uint4 v7 = s4[i + 448];
+        d4[i] = v0;
+        d4[i + 64] = v1;
+        d4[i + 128] = v2;
+        d4[i + 192] = v3;
+        d4[i + 256] = v4;
+        d4[i + 320] = v5;
+        d4[i + 384] = v6;
+        d4[i + 448] = v7;
+      }
       for (; i + 192 < n4; i += 256) {
         uint4 a = s4[i];
         uint4 b = s4[i + 64];
