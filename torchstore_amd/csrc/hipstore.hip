@@ -559,6 +559,20 @@ static void cast_copy(uintptr_t src, int src_dtype, uintptr_t dst,
 // host pinning
 // ---------------------------------------------------------------------------
 
+// single UVA-resolved copy on a dedicated pool stream (blocking).  Used by
+// the SHM transport for D2H staging: torch's copy_ into externally
+// registered host memory runs ~2x slower than a direct hipMemcpyAsync.
+static void sdma_copy(uintptr_t dst, uintptr_t src, uint64_t nbytes,
+                      int device) {
+  DevicePool& p = pool_for(device);
+  HIP_CHECK(hipSetDevice(device));
+  hipStream_t s = p.streams[1];
+  HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(dst),
+                           reinterpret_cast<void*>(src), nbytes,
+                           hipMemcpyDefault, s));
+  HIP_CHECK(hipStreamSynchronize(s));
+}
+
 static void host_register(uintptr_t ptr, uint64_t nbytes) {
   HIP_CHECK(hipHostRegister(reinterpret_cast<void*>(ptr), nbytes,
                             hipHostRegisterPortable));
@@ -596,6 +610,9 @@ PYBIND11_MODULE(_hipstore, m) {
   m.def("cast_copy", &cast_copy, py::arg("src"), py::arg("src_dtype"),
         py::arg("dst"), py::arg("dst_dtype"), py::arg("numel"),
         py::arg("device"), py::arg("stream"));
+  m.def("sdma_copy", &sdma_copy, py::arg("dst"), py::arg("src"),
+        py::arg("nbytes"), py::arg("device"),
+        py::call_guard<py::gil_scoped_release>());
   m.def("host_register", &host_register);
   m.def("host_unregister", &host_unregister);
   m.def("device_count", &device_count);
