@@ -17,7 +17,9 @@ staging copy and the cast are one kernel pass.
 
 from __future__ import annotations
 
-from typing import Any, Dict, Optional
+import asyncio
+import os
+from typing import Any, Dict, List, Optional
 
 import torch
 
@@ -27,6 +29,18 @@ from torchstore_amd.utils.logging import LatencyTracker, get_logger
 logger = get_logger("torchstore_amd.state_dict")
 
 MAPPING_KEY = "<MAPPING>"
+
+# pipeline factor: big state_dicts move as N concurrent sub-batches so RPC
+# framing/parsing overlaps the volume's bulk copies
+_PIPELINE = max(1, int(os.environ.get("TORCHSTORE_AMD_SD_PIPELINE", "4")))
+
+
+def _split(d: Dict[str, Any], n: int) -> List[Dict[str, Any]]:
+    if n <= 1 or len(d) <= 8:
+        return [d]
+    items = list(d.items())
+    size = (len(items) + n - 1) // n
+    return [dict(items[i : i + size]) for i in range(0, len(items), size)]
 
 
 def _flatten(state_dict: Dict[str, Any]):
@@ -96,7 +110,10 @@ async def put_state_dict(
     tracker.step("flatten")
     flat = _cast_floating(flat, transfer_dtype)
     tracker.step("cast")
-    await client.put_batch({f"{key}/{k}": v for k, v in flat.items()})
+    prefixed = {f"{key}/{k}": v for k, v in flat.items()}
+    await asyncio.gather(
+        *(client.put_batch(chunk) for chunk in _split(prefixed, _PIPELINE))
+    )
     tracker.step("put_batch", _nbytes(flat))
     # commit marker: written last, fetched first by readers
     await client.put(f"{key}/{MAPPING_KEY}", mapping)
@@ -145,7 +162,12 @@ async def get_state_dict(
                     f"{sorted(missing)[:5]}..."
                 )
         fetches = {f"{key}/{k}": v for k, v in user_flat.items()}
-        results = await client.get_batch(fetches)
+        chunks = await asyncio.gather(
+            *(client.get_batch(c) for c in _split(fetches, _PIPELINE))
+        )
+        results = {}
+        for c in chunks:
+            results.update(c)
         tracker.step("get_batch", _nbytes(user_flat))
         flat = {k: results[f"{key}/{k}"] for k in user_flat}
         out = _unflatten(flat, user_mapping)
@@ -154,7 +176,12 @@ async def get_state_dict(
 
     flat_keys = list(mapping.keys())
     fetches = {f"{key}/{k}": None for k in flat_keys}
-    results = await client.get_batch(fetches)
+    chunks = await asyncio.gather(
+        *(client.get_batch(c) for c in _split(fetches, _PIPELINE))
+    )
+    results = {}
+    for c in chunks:
+        results.update(c)
     tracker.step("get_batch")
     flat = {k: results[f"{key}/{k}"] for k in flat_keys}
     out = _unflatten(flat, mapping)
