@@ -15,6 +15,9 @@ import torch
 def cast_tensor(t: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     if t.dtype == dtype:
         return t
+    if type(t) is not torch.Tensor:
+        # tensor subclasses (DTensor etc.) go through their own dispatch
+        return t.to(dtype)
     if t.device.type == "cuda":
         from torchstore_amd.ops import gpu
 
