@@ -193,6 +193,90 @@ async def test_two_volumes_one_gpu_fanout():
         await ts.shutdown()
 
 
+from torchstore_amd.runtime import Actor, endpoint  # noqa: E402
+
+
+class WeightSource(Actor):
+    """Weight-sync trainer stand-in (module scope for spawn pickling)."""
+
+    def __init__(self, controller):
+        import os
+
+        os.environ["RANK"] = "0"
+        torch.cuda.set_device(0)
+        ts.attach(controller, SingletonStrategy())
+        torch.manual_seed(7)
+        self.params = {
+            "w": torch.randn(512, 512, device="cuda"),
+            "b": torch.randn(512, device="cuda"),
+            "master": torch.randn(256, 256, device="cuda", dtype=torch.float32),
+        }
+
+    @endpoint
+    async def push(self, transfer_dtype=None):
+        await ts.put_state_dict(
+            self.params, "dsync", direct=True,
+            transfer_dtype=transfer_dtype, rank=0, world_size=1,
+        )
+        return "ok"
+
+    @endpoint
+    def mutate(self):
+        with torch.no_grad():
+            self.params["w"].add_(1.0)
+        torch.cuda.synchronize()
+        return "ok"
+
+    @endpoint
+    def snapshot(self):
+        return {k: v.cpu() for k, v in self.params.items()}
+
+
+@requires_gpu
+async def test_direct_weight_sync_real_ipc():
+    """Trainer process exports live param handles; this process pulls
+    one-sided over real HIP IPC; in-place updates visible on re-pull."""
+    from torchstore_amd.runtime import spawn_actors
+
+    controller = await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(),
+        storage_device="auto",
+    )
+    mesh = None
+    try:
+        import asyncio as aio
+
+        mesh = await aio.to_thread(
+            spawn_actors, 1, WeightSource, "wsource", controller
+        )
+        src = mesh.handles[0]
+        assert await src.push.call_one() == "ok"
+
+        dst = {
+            "w": torch.zeros(512, 512, device="cuda"),
+            "b": torch.zeros(512, device="cuda"),
+            "master": torch.zeros(256, 256, device="cuda", dtype=torch.float32),
+        }
+        await ts.get_state_dict("dsync", dst, direct=True)
+        torch.cuda.synchronize()
+        snap = await src.snapshot.call_one()
+        for k in dst:
+            assert torch.equal(dst[k].cpu(), snap[k]), k
+
+        # optimizer-style in-place update: visible WITHOUT re-registration
+        await src.mutate.call_one()
+        await ts.get_state_dict("dsync", dst, direct=True)
+        torch.cuda.synchronize()
+        snap2 = await src.snapshot.call_one()
+        assert torch.equal(dst["w"].cpu(), snap2["w"])
+        assert not torch.equal(snap["w"], snap2["w"])
+    finally:
+        if mesh is not None:
+            await mesh.stop()
+        await ts.shutdown()
+
+
 @requires_gpu
 async def test_gpu_put_does_not_sync_foreign_stream():
     """The reference's stream-isolation invariant: a put must not wait on

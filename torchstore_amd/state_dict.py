@@ -32,7 +32,7 @@ MAPPING_KEY = "<MAPPING>"
 
 # pipeline factor: big state_dicts move as N concurrent sub-batches so RPC
 # framing/parsing overlaps the volume's bulk copies
-_PIPELINE = max(1, int(os.environ.get("TORCHSTORE_AMD_SD_PIPELINE", "4")))
+_PIPELINE = max(1, int(os.environ.get("TORCHSTORE_AMD_SD_PIPELINE", "8")))
 
 
 def _split(d: Dict[str, Any], n: int) -> List[Dict[str, Any]]:
