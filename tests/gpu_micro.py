@@ -96,6 +96,17 @@ def main():
     )
     print(f"  bw (rd+wr): {nb/dt/1e9:.1f} GB/s")
 
+    # SDMA pitched copy of the same strided pack
+    dt = timeit(
+        "hipMemcpy2D same pack",
+        lambda: e.copy_batch_2d([
+            (out.data_ptr(), 0, 4096 * 4,
+             view.data_ptr(), 0, 8192 * 4, 4096 * 4, 8192)
+        ]),
+        n=10,
+    )
+    print(f"  bw (rd+wr): {nb/dt/1e9:.1f} GB/s")
+
     # K2 batched many-slice scatter vs a torch copy loop (the reshard case)
     n_slices = 256
     srcs2 = [torch.randn(512, 512, device="cuda", dtype=torch.bfloat16)
