@@ -10,7 +10,7 @@ cached LocalClient per process.
 from __future__ import annotations
 
 import asyncio
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List, Optional, Sequence
 
 from torchstore_amd.client import LocalClient

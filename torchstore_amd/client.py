@@ -14,7 +14,7 @@ Responsibilities (reference: torchstore ``client.py``):
 from __future__ import annotations
 
 import asyncio
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 import torch
@@ -25,7 +25,6 @@ from torchstore_amd.runtime import ActorHandle
 from torchstore_amd.storage import OBJ_SENTINEL, TensorMeta
 from torchstore_amd.strategy import (
     PlacementStrategy,
-    SingletonStrategy,
     StorageVolumeRef,
     strategy_from_spec,
 )

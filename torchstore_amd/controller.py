@@ -19,7 +19,7 @@ from dataclasses import dataclass, field
 from enum import Enum
 from typing import Dict, List, Optional, Sequence, Set, Tuple
 
-from torchstore_amd.runtime import Actor, ActorHandle, ActorMesh, endpoint
+from torchstore_amd.runtime import Actor, ActorHandle, endpoint
 from torchstore_amd.types import Request, TensorSlice
 from torchstore_amd.utils.logging import get_logger
 from torchstore_amd.utils.trie import Trie

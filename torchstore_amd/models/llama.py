@@ -7,7 +7,7 @@ shard-by-shard so each rank only allocates its own slice.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, Optional, Tuple
 
 import torch
 

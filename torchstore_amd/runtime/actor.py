@@ -20,7 +20,7 @@ from __future__ import annotations
 import asyncio
 import weakref
 from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional, Sequence, Tuple
+from typing import Any, List, Sequence, Tuple
 
 from torchstore_amd.runtime.rpc import (
     SHUTDOWN_METHOD,

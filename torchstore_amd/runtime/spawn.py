@@ -18,7 +18,7 @@ import asyncio
 import multiprocessing as mp
 import os
 from dataclasses import dataclass
-from typing import Any, Dict, Optional, Sequence, Tuple, Type
+from typing import Dict, Optional, Sequence, Tuple, Type
 
 from torchstore_amd.runtime.actor import Actor, ActorHandle, ActorMesh
 from torchstore_amd.runtime.rpc import RpcServer

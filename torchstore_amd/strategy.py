@@ -17,10 +17,9 @@ from __future__ import annotations
 
 import os
 import socket
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from dataclasses import dataclass
+from typing import List, Optional
 
-from torchstore_amd.controller import VolumeInfo
 from torchstore_amd.runtime import ActorHandle
 from torchstore_amd.transport.base import TransportContext, TransportType
 

@@ -17,8 +17,8 @@ Design notes (MI355X build):
 from __future__ import annotations
 
 import itertools
-from dataclasses import dataclass, field, replace
-from typing import Any, Dict, Optional, Tuple
+from dataclasses import dataclass, replace
+from typing import Any, Optional, Tuple
 
 import torch
 

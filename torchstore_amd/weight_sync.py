@@ -27,7 +27,7 @@ logic runs in CPU tests — mirroring the reference's MockRDMABuffer tests.
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 import torch
