@@ -217,7 +217,16 @@ static void copy_batch_2d(
 // global unit list and binary-search their slice in a prefix array.
 
 static constexpr int kMaxDims = 7;       // outer dims (innermost is the row)
-static constexpr uint32_t kTileBytes = 16384;  // 256 threads * 16B * 4 iters
+static uint32_t tile_bytes_cfg() {
+  static uint32_t v = [] {
+    const char* e = getenv("HIPSTORE_TILE");
+    uint32_t t = e ? (uint32_t)atoi(e) : 65536;
+    if (t < 4096) t = 4096;
+    return t;
+  }();
+  return v;
+}
+#define kTileBytes tile_bytes_cfg()
 
 struct SliceDesc {
   uintptr_t src;
@@ -252,7 +261,7 @@ __device__ __forceinline__ void row_offsets(const SliceDesc& d, uint64_t row,
 // per lane (64 lanes x 16B x 4 = 4 KiB outstanding per wave).
 __global__ void __launch_bounds__(256)
 copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
-                   uint64_t total_units) {
+                   uint64_t total_units, uint32_t tile_bytes) {
   const uint32_t wave = threadIdx.x >> 6;
   const uint32_t lane = threadIdx.x & 63u;
   const uint64_t stride = (uint64_t)gridDim.x * 4;
@@ -270,8 +279,8 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
     uint32_t tile = (uint32_t)(local % d.tiles_per_row);
     int64_t soff, doff;
     row_offsets(d, row, soff, doff);
-    uint32_t start = tile * kTileBytes;
-    uint32_t len = min(kTileBytes, d.row_bytes - start);
+    uint32_t start = tile * tile_bytes;
+    uint32_t len = min(tile_bytes, d.row_bytes - start);
     const char* src = reinterpret_cast<const char*>(d.src) + soff + start;
     char* dst = reinterpret_cast<char*>(d.dst) + doff + start;
     uintptr_t sa = reinterpret_cast<uintptr_t>(src);
@@ -352,7 +361,7 @@ static void launch_slice_descs(std::vector<SliceDesc>& descs, uint64_t units,
   uint32_t grid = (uint32_t)std::min<uint64_t>((units + 3) / 4, 2048);
   hipLaunchKernelGGL(copy_slices_kernel, dim3(grid), dim3(256), 0, stream,
                      reinterpret_cast<const SliceDesc*>(p.d_desc),
-                     (uint32_t)descs.size(), units);
+                     (uint32_t)descs.size(), units, kTileBytes);
   HIP_CHECK(hipGetLastError());
   HIP_CHECK(hipEventRecord(p.desc_evt, stream));
 }
