@@ -217,16 +217,17 @@ static void copy_batch_2d(
 // global unit list and binary-search their slice in a prefix array.
 
 static constexpr int kMaxDims = 7;       // outer dims (innermost is the row)
-static uint32_t tile_bytes_cfg() {
-  static uint32_t v = [] {
+// tile size per launch: big batches amortize per-tile overhead with 128 KiB
+// tiles; smaller ones keep 16 KiB tiles so enough waves stay busy
+// (HIPSTORE_TILE overrides for experiments)
+static uint32_t pick_tile(uint64_t total_bytes) {
+  static uint32_t forced = [] {
     const char* e = getenv("HIPSTORE_TILE");
-    uint32_t t = e ? (uint32_t)atoi(e) : 65536;
-    if (t < 4096) t = 4096;
-    return t;
+    return e ? (uint32_t)atoi(e) : 0u;
   }();
-  return v;
+  if (forced >= 4096) return forced;
+  return total_bytes > (512ull << 20) ? 131072u : 16384u;
 }
-#define kTileBytes tile_bytes_cfg()
 
 struct SliceDesc {
   uintptr_t src;
@@ -339,7 +340,8 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
 
 // upload descriptors through the pinned staging buffer + launch the kernel
 static void launch_slice_descs(std::vector<SliceDesc>& descs, uint64_t units,
-                               int device, hipStream_t stream) {
+                               int device, hipStream_t stream,
+                               uint32_t tile) {
   if (units == 0 || descs.empty()) return;
   DevicePool& p = pool_for(device);
   HIP_CHECK(hipSetDevice(device));
@@ -361,21 +363,21 @@ static void launch_slice_descs(std::vector<SliceDesc>& descs, uint64_t units,
   uint32_t grid = (uint32_t)std::min<uint64_t>((units + 3) / 4, 2048);
   hipLaunchKernelGGL(copy_slices_kernel, dim3(grid), dim3(256), 0, stream,
                      reinterpret_cast<const SliceDesc*>(p.d_desc),
-                     (uint32_t)descs.size(), units, kTileBytes);
+                     (uint32_t)descs.size(), units, tile);
   HIP_CHECK(hipGetLastError());
   HIP_CHECK(hipEventRecord(p.desc_evt, stream));
 }
 
 // trivial desc for a flat byte copy (used by copy_batch's same-device path)
 static SliceDesc flat_desc(uintptr_t dst, uintptr_t src, uint64_t nbytes,
-                           uint64_t units_prefix) {
+                           uint64_t units_prefix, uint32_t tile) {
   SliceDesc d{};
   d.src = src;
   d.dst = dst;
   d.rows = 1;
   d.units_prefix = units_prefix;
   d.row_bytes = (uint32_t)nbytes;
-  d.tiles_per_row = (uint32_t)((nbytes + kTileBytes - 1) / kTileBytes);
+  d.tiles_per_row = (uint32_t)((nbytes + tile - 1) / tile);
   if (d.tiles_per_row == 0) d.tiles_per_row = 1;
   d.ndim = 0;
   return d;
@@ -386,8 +388,8 @@ static void copy_batch(
         copies) {
   if (copies.empty()) return;
   // group same-device copies per device; cross-device go via SDMA
-  std::unordered_map<int, std::vector<SliceDesc>> per_device;
-  std::unordered_map<int, uint64_t> per_device_units;
+  std::unordered_map<int, std::vector<std::tuple<uintptr_t, uintptr_t, uint64_t>>>
+      flat_per_device;
   std::vector<hipStream_t> used;
   int i = 0;
   for (const auto& c : copies) {
@@ -398,9 +400,7 @@ static void copy_batch(
     uint64_t n = std::get<4>(c);
     if (n == 0) continue;
     if (dst_dev == src_dev && n <= UINT32_MAX) {
-      uint64_t& units = per_device_units[dst_dev];
-      per_device[dst_dev].push_back(flat_desc(dst, src, n, units));
-      units += per_device[dst_dev].back().tiles_per_row;
+      flat_per_device[dst_dev].emplace_back(dst, src, n);
       continue;
     }
     DevicePool& p = pool_for(dst_dev);
@@ -415,11 +415,22 @@ static void copy_batch(
     (void)src_dev;
     used.push_back(s);
   }
-  for (auto& kv : per_device) {
+  for (auto& kv : flat_per_device) {
     int device = kv.first;
+    uint64_t total = 0;
+    for (auto& t : kv.second) total += std::get<2>(t);
+    uint32_t tile = pick_tile(total);
+    std::vector<SliceDesc> descs;
+    descs.reserve(kv.second.size());
+    uint64_t units = 0;
+    for (auto& t : kv.second) {
+      descs.push_back(flat_desc(std::get<0>(t), std::get<1>(t),
+                                std::get<2>(t), units, tile));
+      units += descs.back().tiles_per_row;
+    }
     DevicePool& p = pool_for(device);
     hipStream_t s = p.streams[0];
-    launch_slice_descs(kv.second, per_device_units[device], device, s);
+    launch_slice_descs(descs, units, device, s, tile);
     used.push_back(s);
   }
   for (hipStream_t s : used) HIP_CHECK(hipStreamSynchronize(s));
@@ -435,6 +446,13 @@ static void copy_slices(const std::vector<PySlice>& slices, int device,
                         uintptr_t stream_handle, bool blocking) {
   if (slices.empty()) return;
   size_t n = slices.size();
+  uint64_t total_bytes = 0;
+  for (const auto& s : slices) {
+    uint64_t rows = 1;
+    for (uint64_t d : std::get<3>(s)) rows *= d;
+    total_bytes += rows * std::get<2>(s);
+  }
+  uint32_t tile = pick_tile(total_bytes);
   std::vector<SliceDesc> descs(n);
   uint64_t units = 0;
   for (size_t i = 0; i < n; ++i) {
@@ -457,7 +475,7 @@ static void copy_slices(const std::vector<PySlice>& slices, int device,
       d.rows *= shape[k];
     }
     d.row_bytes = (uint32_t)row_bytes;
-    d.tiles_per_row = (uint32_t)((row_bytes + kTileBytes - 1) / kTileBytes);
+    d.tiles_per_row = (uint32_t)((row_bytes + tile - 1) / tile);
     if (d.tiles_per_row == 0) d.tiles_per_row = 1;
     d.units_prefix = units;
     units += d.rows * d.tiles_per_row;
@@ -465,7 +483,7 @@ static void copy_slices(const std::vector<PySlice>& slices, int device,
   if (units == 0) return;
 
   launch_slice_descs(descs, units, device,
-                     reinterpret_cast<hipStream_t>(stream_handle));
+                     reinterpret_cast<hipStream_t>(stream_handle), tile);
   if (blocking) {
     HIP_CHECK(
         hipStreamSynchronize(reinterpret_cast<hipStream_t>(stream_handle)));
