@@ -102,6 +102,29 @@ async def test_gpu_slice_fetch():
 
 
 @requires_gpu
+async def test_gpu_large_strided_slice_fetch():
+    """A >32MB strided slice of a stored tensor: exercises the fused
+    SDMA pitched-read path in volume_send."""
+    from torchstore_amd.types import LocalShard, TensorSlice
+
+    async def body():
+        t = torch.randn(8192, 8192, device="cuda")  # 256 MB
+        await ts.put("big", t)
+        dest = LocalShard(
+            tensor=torch.zeros(8192, 4096, device="cuda"),
+            slice=TensorSlice(
+                offsets=(0, 2048), local_shape=(8192, 4096),
+                global_shape=(8192, 8192), coordinates=(), mesh_shape=(),
+            ),
+        )
+        await ts.get("big", dest)
+        torch.cuda.synchronize()
+        assert torch.equal(dest.tensor, t[:, 2048:6144])
+
+    await _with_store(body, transport=TransportType.HIP_IPC)
+
+
+@requires_gpu
 async def test_gpu_shm_transport():
     """SHM path with GPU tensors: pinned staging + copy streams."""
 

@@ -116,6 +116,16 @@ async def test_tensor_rpc_and_exceptions():
         await close_connections()
 
 
+class FailingActor(Actor):
+    def __init__(self):
+        raise ValueError("deliberate ctor failure")
+
+
+def test_spawn_failure_reports_traceback():
+    with pytest.raises(RuntimeError, match="deliberate ctor failure"):
+        spawn_actors(1, FailingActor, "fails")
+
+
 async def test_mesh_slice_and_concurrent_calls():
     mesh = spawn_actors(4, Echo, "echo-m", mesh_shape=(2, 2))
     try:
