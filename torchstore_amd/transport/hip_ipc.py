@@ -399,6 +399,7 @@ class HipIpcTransportBuffer(TransportBuffer):
         reply: List[Tuple[str, Any]] = []
         fused: List[Tuple[torch.Tensor, int]] = []   # same-device: K1 writes
         copies: List[Tuple[int, int, int, int, int]] = []  # cross-device SDMA
+        copies_2d: List[Tuple[int, int, int, int, int, int, int, int]] = []
         device = None
         for (kind, value), r, v in zip(self.payload, requests, values):
             if kind == "fetch_obj" or not isinstance(v, torch.Tensor):
@@ -425,12 +426,27 @@ class HipIpcTransportBuffer(TransportBuffer):
                 # client's mapped destination — no packed intermediate
                 fused.append((v, dst_ptr))
             else:
-                vc = gpu_ops.pack_region(v)  # K1 gather for strided views
-                copies.append(
-                    (dst_ptr, desc.device_index, vc.data_ptr(),
-                     v.device.index, desc.nbytes)
+                # cross-device (over xGMI): large 2-D strided sources go
+                # through one SDMA pitched read — still no packed
+                # intermediate; only irregular layouts pack first
+                sp = (
+                    gpu_ops._pitched_params(v)
+                    if desc.nbytes >= gpu_ops._SDMA_2D_MIN_BYTES
+                    and not v.is_contiguous()
+                    else None
                 )
-                self._hold.append(vc)
+                if sp is not None:
+                    copies_2d.append(
+                        (dst_ptr, v.device.index, sp[1],
+                         v.data_ptr(), v.device.index, sp[0], sp[1], sp[2])
+                    )
+                else:
+                    vc = gpu_ops.pack_region(v)  # K1 gather for strided views
+                    copies.append(
+                        (dst_ptr, desc.device_index, vc.data_ptr(),
+                         v.device.index, desc.nbytes)
+                    )
+                    self._hold.append(vc)
             reply.append(("done", None))
         if fused:
             import asyncio
@@ -447,13 +463,16 @@ class HipIpcTransportBuffer(TransportBuffer):
                      vc.numel() * vc.element_size())
                 )
                 self._hold.append(vc)
-        if copies:
-            # K1 pack kernels ran on the current stream; the pool streams
-            # used by copy_batch must observe their writes
-            torch.cuda.current_stream(device).synchronize()
+        if copies or copies_2d:
             import asyncio
 
-            await asyncio.to_thread(_run_copies, copies)
+            if copies:
+                # K1 pack kernels ran on the current stream; the pool
+                # streams used by copy_batch must observe their writes
+                torch.cuda.current_stream(device).synchronize()
+                await asyncio.to_thread(_run_copies, copies)
+            if copies_2d:
+                await asyncio.to_thread(_ext().copy_batch_2d, copies_2d)
         return reply
 
     def client_complete_get(self, requests, reply) -> List[Any]:
