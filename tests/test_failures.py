@@ -64,6 +64,20 @@ async def test_reset_clears_everything():
         await ts.shutdown()
 
 
+async def test_controller_stats():
+    await ts.initialize(
+        num_storage_volumes=1, strategy=SingletonStrategy(), storage_device="cpu"
+    )
+    try:
+        await ts.put("a", torch.ones(4))
+        await ts.put("b", {"x": 1})
+        session = ts.api._sessions["default"]
+        stats = await session.controller.stats.call_one()
+        assert stats["keys"] == 2 and stats["volumes"] == 1
+    finally:
+        await ts.shutdown()
+
+
 async def test_double_shutdown_is_safe():
     await ts.initialize(
         num_storage_volumes=1, strategy=SingletonStrategy(), storage_device="cpu"

@@ -185,6 +185,25 @@ class Controller(Actor):
         return locations is not None and self._is_fully_committed(locations)
 
     @endpoint
+    def stats(self) -> Dict[str, int]:
+        """Index-level observability: key/volume/shard counts."""
+        n_keys = 0
+        n_sharded = 0
+        n_shards = 0
+        for key in self.index:
+            n_keys += 1
+            for info in self.index[key].values():
+                if info.object_type == ObjectType.TENSOR_SLICE:
+                    n_sharded += 1
+                    n_shards += len(info.tensor_slices)
+        return {
+            "keys": n_keys,
+            "volumes": len(self.volumes),
+            "sharded_entries": n_sharded,
+            "total_shards": n_shards,
+        }
+
+    @endpoint
     async def teardown(self) -> None:
         """Reset every volume's storage (volume processes stay up)."""
         for v in self.volumes.values():
