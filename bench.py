@@ -5,8 +5,14 @@ FSDP-style Shard(0) layout of a random-init Llama-3-8B (~16.06 GB bf16),
 one **step** = push the full state_dict into GPU-resident storage volumes
 (put_state_dict) + pull it back in the TP-style layout (get_state_dict with
 resharding).  value = whole-job aggregate GB/s moved (put+get bytes / step
-time, max over ranks).  The reference (meta-pytorch/torchstore) publishes
-no numbers — this is the self-measured baseline.
+time, max over ranks).  ``--mode direct`` instead measures the one-sided
+weight-sync pull (a generator process reads the trainer's live parameters
+over xGMI).  The reference (meta-pytorch/torchstore) publishes no numbers
+ — this is the self-measured baseline.
+
+Shards are expressed as :class:`LocalShard` (explicit TensorSlice layouts),
+so the bench needs no process group: ranks coordinate through a TCPStore
+(barriers + max-reduce).  torchrun only supplies the env.
 
 Single GPU:      python bench.py --steps 5 --warmup 2
 N GPUs (driver): python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
@@ -74,41 +80,62 @@ def parse_args():
     return p.parse_args()
 
 
-def setup_dist(args):
-    rank = int(os.environ.get("RANK", "0"))
-    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
-    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
-    if world > 1:
-        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29531")
-        torch.distributed.init_process_group(
-            "nccl", rank=rank, world_size=world
-        )
-        torch.cuda.set_device(local_rank)
-    else:
-        if torch.cuda.is_available():
-            torch.cuda.set_device(0)
-    return rank, world, local_rank
+class Coord:
+    """Rank coordination over a TCPStore: barriers + gather (no PG needed)."""
+
+    def __init__(self, rank: int, world: int):
+        self.rank = rank
+        self.world = world
+        self._n = 0
+        self.store = None
+        if world > 1:
+            from datetime import timedelta
+
+            from torch.distributed import TCPStore
+
+            self.store = TCPStore(
+                os.environ.get("MASTER_ADDR", "127.0.0.1"),
+                int(os.environ.get("MASTER_PORT", "29500")) + 2,
+                world, is_master=rank == 0,
+                timeout=timedelta(seconds=300),
+                wait_for_workers=True,
+            )
+
+    def barrier(self):
+        if self.store is None:
+            return
+        self._n += 1
+        key = f"bar/{self._n}"
+        self.store.add(key, 1)
+        while int(self.store.add(key, 0)) < self.world:
+            time.sleep(0.001)
+
+    def gather_floats(self, value: float):
+        """Every rank contributes; rank 0 gets the list, others None."""
+        if self.store is None:
+            return [value]
+        self._n += 1
+        self.store.set(f"g/{self._n}/{self.rank}", repr(float(value)))
+        self.barrier()
+        if self.rank != 0:
+            return None
+        return [
+            float(self.store.get(f"g/{self._n - 1}/{r}").decode())
+            for r in range(self.world)
+        ]
 
 
-async def run_bench(args, rank, world, local_rank):
+async def run_bench(args, rank, world, local_rank, coord: Coord):
     import torchstore_amd as ts
     from torchstore_amd.models import llama
     from torchstore_amd.strategy import LocalRankStrategy
+    from torchstore_amd.types import LocalShard
 
     device = f"cuda:{local_rank}"
-    mesh = None
-    if world > 1:
-        from torch.distributed.device_mesh import init_device_mesh
-
-        mesh = init_device_mesh("cuda", (world,))
-
     layers = args.layers or llama.LAYERS
-    src_sd = llama.make_sharded_state_dict(
-        mesh, llama.fsdp_placement, device=device, layers=layers, seed=1234 + rank
-    )
-    dst_sd = llama.make_sharded_state_dict(
-        mesh, llama.tp_placement, device=device, layers=layers, zero=True
+    src_sd = llama.make_local_shard_state_dict(
+        rank, world, llama.fsdp_placement, device=device, layers=layers,
+        zero=False,
     )
     shapes = llama.llama3_8b_shapes(layers)
     payload_bytes = llama.total_bytes(shapes, torch.bfloat16)
@@ -127,10 +154,11 @@ async def run_bench(args, rank, world, local_rank):
         )
 
     def barrier():
-        if world > 1:
-            torch.distributed.barrier()
+        coord.barrier()
         torch.cuda.synchronize()
 
+    gen_mesh = None
+    pull_bytes = [0]
     if args.mode == "direct":
         # trainer = this process; generator = a separate process on the same
         # GPU (IPC handles cannot be opened by their exporting process)
@@ -139,14 +167,12 @@ async def run_bench(args, rank, world, local_rank):
         await ts.put_state_dict(
             src_sd, "bench", direct=True, rank=rank, world_size=world
         )
-        if world > 1:
-            torch.distributed.barrier()
+        coord.barrier()
         gen_mesh = await asyncio.to_thread(
             spawn_actors, 1, GeneratorActor, f"generator-{rank}",
             controller, "bench", layers, world, rank, local_rank,
         )
         gen = gen_mesh.handles[0]
-        pull_bytes = [0]
 
         async def one_step():
             # push = staging refresh (no cast here → sync only);
@@ -155,30 +181,33 @@ async def run_bench(args, rank, world, local_rank):
             _dt, pull_bytes[0] = await gen.pull.call_one()
 
     else:
+        dst_sd = llama.make_local_shard_state_dict(
+            rank, world, llama.tp_placement, device=device, layers=layers,
+        )
 
         async def one_step():
             await ts.put_state_dict(src_sd, "bench")
-            if world > 1:
-                torch.distributed.barrier()  # all shards committed
+            coord.barrier()  # all shards committed before the reshard pull
             await ts.get_state_dict("bench", dst_sd)
 
     for _ in range(args.warmup):
         await one_step()
     barrier()
+
     # correctness guard (outside the timed region): the norm weight has the
     # same layout in both placements, so the pulled local shard must equal
     # the pushed one bit-for-bit
     probe = "model.norm.weight"
-    src_t = src_sd[probe]
-    put_local = getattr(src_t, "to_local", lambda: src_t)
+
+    def _local(x):
+        return x.tensor if isinstance(x, LocalShard) else x
+
     if args.mode == "direct":
-        ok = await gen.verify.call_one(probe, put_local().cpu())
+        ok = await gen.verify.call_one(probe, _local(src_sd[probe]).cpu())
         if not ok:
             raise RuntimeError("bench correctness probe failed (direct pull)")
     else:
-        dst_t = dst_sd[probe]
-        get_local = getattr(dst_t, "to_local", lambda: dst_t)
-        if not torch.equal(get_local(), put_local()):
+        if not torch.equal(_local(dst_sd[probe]), _local(src_sd[probe])):
             raise RuntimeError("bench correctness probe failed: pulled != pushed")
 
     t0 = time.perf_counter()
@@ -187,24 +216,20 @@ async def run_bench(args, rank, world, local_rank):
     barrier()
     elapsed = time.perf_counter() - t0
 
-    # max over ranks
-    if world > 1:
-        t = torch.tensor([elapsed], device=device)
-        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
-        elapsed = t.item()
-
-    ms_per_step = elapsed / args.steps * 1e3
+    times = coord.gather_floats(elapsed)
     if args.mode == "direct":
-        moved = pull_bytes[0]  # actual bytes read per pull, this rank
-        if world > 1:
-            t = torch.tensor([float(moved)], device=device)
-            torch.distributed.all_reduce(t)
-            moved = int(t.item())
+        rank_bytes = coord.gather_floats(float(pull_bytes[0]))
     else:
-        moved = 2 * payload_bytes  # put + get per step, whole job
-    gbps = moved / (elapsed / args.steps) / 1e9
+        rank_bytes = None
 
     if rank == 0:
+        elapsed = max(times)
+        ms_per_step = elapsed / args.steps * 1e3
+        if args.mode == "direct":
+            moved = int(sum(rank_bytes))  # actual bytes read per pull, all ranks
+        else:
+            moved = 2 * payload_bytes  # put + get per step, whole job
+        gbps = moved / (elapsed / args.steps) / 1e9
         print(json.dumps({
             "metric": "llama3_8b_state_dict_sync_GBps",
             "value": round(gbps, 2),
@@ -226,24 +251,28 @@ async def run_bench(args, rank, world, local_rank):
                 "payload_gb": round(payload_bytes / 1e9, 2),
                 "mode": args.mode,
             },
-        }))
+        }), flush=True)
 
-    if args.mode == "direct":
+    if gen_mesh is not None:
         await gen_mesh.stop()
-    if world > 1:
-        torch.distributed.barrier()
+    coord.barrier()
     await ts.shutdown()  # collective in SPMD mode
-    if world > 1:
-        torch.distributed.destroy_process_group()
 
 
 def main():
     args = parse_args()
     if not torch.cuda.is_available():
         raise SystemExit("bench.py needs MI355X GPUs (torch.cuda unavailable)")
-    rank, world, local_rank = setup_dist(args)
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    torch.cuda.set_device(local_rank % torch.cuda.device_count())
+    coord = Coord(rank, world)
     try:
-        asyncio.run(run_bench(args, rank, world, local_rank))
+        asyncio.run(
+            run_bench(args, rank, world,
+                      local_rank % torch.cuda.device_count(), coord)
+        )
     except Exception:
         import sys
         import traceback
