@@ -125,6 +125,44 @@ async def test_gpu_large_strided_slice_fetch():
 
 
 @requires_gpu
+async def test_bounce_coalesced_get(monkeypatch):
+    """The cross-device bounce path (pack -> one SDMA -> scatter), forced
+    on one GPU by faking the volume's device index."""
+    from torchstore_amd.transport.hip_ipc import HipIpcTransportBuffer
+
+    monkeypatch.setattr(
+        HipIpcTransportBuffer, "_volume_device_index", lambda self: 7
+    )
+
+    async def body():
+        items = {f"p{i}": torch.randn(256, 256, device="cuda") for i in range(12)}
+        await ts.put_batch(items)
+        dests = {k: torch.zeros_like(v) for k, v in items.items()}
+        out = await ts.get_batch(dests)
+        torch.cuda.synchronize()
+        for k, v in items.items():
+            assert torch.equal(out[k], v), k
+
+        # strided dest (region fetch) through the bounce path
+        from torchstore_amd.types import LocalShard, TensorSlice
+
+        t = torch.randn(512, 512, device="cuda")
+        await ts.put("bigp", t)
+        dest = LocalShard(
+            tensor=torch.zeros(512, 256, device="cuda"),
+            slice=TensorSlice(
+                offsets=(0, 128), local_shape=(512, 256),
+                global_shape=(512, 512), coordinates=(), mesh_shape=(),
+            ),
+        )
+        await ts.get("bigp", dest)
+        torch.cuda.synchronize()
+        assert torch.equal(dest.tensor, t[:, 128:384])
+
+    await _with_store(body, transport=TransportType.HIP_IPC)
+
+
+@requires_gpu
 async def test_gpu_shm_transport():
     """SHM path with GPU tensors: pinned staging + copy streams."""
 

@@ -35,6 +35,14 @@ MAPPING_KEY = "<MAPPING>"
 _PIPELINE = max(1, int(os.environ.get("TORCHSTORE_AMD_SD_PIPELINE", "8")))
 
 
+async def _get_pipeline(client: LocalClient) -> int:
+    """Gets already fan out one RPC per involved volume; extra client-side
+    splitting only multiplies per-RPC launch/sync overhead once the volume
+    count itself provides the concurrency."""
+    volumes = await client._ensure_volumes()
+    return max(1, _PIPELINE // max(1, len(volumes)))
+
+
 def _split(d: Dict[str, Any], n: int) -> List[Dict[str, Any]]:
     if n <= 1 or len(d) <= 8:
         return [d]
@@ -163,7 +171,10 @@ async def get_state_dict(
                 )
         fetches = {f"{key}/{k}": v for k, v in user_flat.items()}
         chunks = await asyncio.gather(
-            *(client.get_batch(c) for c in _split(fetches, _PIPELINE))
+            *(
+                client.get_batch(c)
+                for c in _split(fetches, await _get_pipeline(client))
+            )
         )
         results = {}
         for c in chunks:
@@ -177,7 +188,10 @@ async def get_state_dict(
     flat_keys = list(mapping.keys())
     fetches = {f"{key}/{k}": None for k in flat_keys}
     chunks = await asyncio.gather(
-        *(client.get_batch(c) for c in _split(fetches, _PIPELINE))
+        *(
+            client.get_batch(c)
+            for c in _split(fetches, await _get_pipeline(client))
+        )
     )
     results = {}
     for c in chunks:

@@ -186,8 +186,10 @@ class HipIpcTransportBuffer(TransportBuffer):
         #   ("ipc", IpcDescriptor) | ("inline", value) | ("chunked", token)
         #   get-side markers: ("fetch_obj"|"fetch_inline", None)
         self.payload: Optional[List[Tuple[str, Any]]] = None
+        self.bounce_descs: List[IpcDescriptor] = []
         self._hold: List[torch.Tensor] = []       # keep exports alive
         self._scratch: Dict[int, torch.Tensor] = {}  # req idx -> dense scratch
+        self._bounces: List[torch.Tensor] = []
 
     def __getstate__(self):
         # local tensor refs must NEVER ride the RPC frame (they would be
@@ -196,6 +198,7 @@ class HipIpcTransportBuffer(TransportBuffer):
         state = super().__getstate__()
         state["_hold"] = []
         state["_scratch"] = {}
+        state["_bounces"] = []
         return state
 
     # -- chunked windows (client side) -----------------------------------
@@ -357,9 +360,25 @@ class HipIpcTransportBuffer(TransportBuffer):
         return out
 
     # ------------------------------------------------------------- get --
+    def _volume_device_index(self) -> int:
+        dev = getattr(self._volume_ref, "device", "") or ""
+        if dev.startswith("cuda"):
+            try:
+                return int(dev.split(":")[1])
+            except (IndexError, ValueError):
+                return 0
+        return -1
+
     async def client_stage_get(self, requests: Sequence[Request]) -> None:
         payload: List[Tuple[str, Any]] = []
         synced: set = set()
+        vol_dev = self._volume_device_index()
+        # cross-device small/strided pieces coalesce through ONE bounce
+        # buffer per op: the volume packs them locally (one kernel), moves
+        # them with ONE SDMA over xGMI, and the client scatters with one K2
+        # launch — instead of a per-piece enqueue storm (the fsdp->tp
+        # reshard makes world^2 pieces per step)
+        bounce_plan: List[Tuple[int, int]] = []  # (request idx, nbytes)
         for i, r in enumerate(requests):
             if r.is_object:
                 payload.append(("fetch_obj", None))
@@ -369,6 +388,12 @@ class HipIpcTransportBuffer(TransportBuffer):
                 raise RuntimeError("IPC get requires pre-allocated destinations")
             if dest.device.type != "cuda":
                 payload.append(("fetch_inline", None))
+                continue
+            nbytes = dest.numel() * dest.element_size()
+            cross = vol_dev >= 0 and vol_dev != dest.device.index
+            if cross and (not dest.is_contiguous() or nbytes < (32 << 20)):
+                payload.append(("bounce", None))  # offsets assigned below
+                bounce_plan.append((i, nbytes))
                 continue
             if dest.is_contiguous():
                 target = dest
@@ -389,6 +414,31 @@ class HipIpcTransportBuffer(TransportBuffer):
                 payload.append(("chunked", token))
             else:
                 payload.append(("ipc", desc))
+
+        self.bounce_descs: List[IpcDescriptor] = []
+        self._bounces: List[torch.Tensor] = []
+        if bounce_plan:
+            device = requests[bounce_plan[0][0]].tensor_val.device
+            cap = 1 << 30  # each bounce buffer stays well under the 2GiB limit
+            aligned = [(i, n, (n + 255) & ~255) for i, n in bounce_plan]
+            remaining = sum(a for _, _, a in aligned)
+            off = 0
+            size = 0
+            b_idx = -1
+            for i, nbytes, a in aligned:
+                if b_idx < 0 or off + a > size:
+                    size = min(cap, max(remaining, a))
+                    buf = torch.empty(size, dtype=torch.uint8, device=device)
+                    self._bounces.append(buf)
+                    self._hold.append(buf)
+                    self.bounce_descs.append(export_tensor(buf))
+                    b_idx += 1
+                    off = 0
+                payload[i] = ("bounce", (b_idx, off))
+                off += a
+                remaining -= a
+            if device.index not in synced:
+                torch.cuda.current_stream(device).synchronize()
         self.payload = payload
 
     async def volume_send(self, requests, values):
@@ -400,6 +450,8 @@ class HipIpcTransportBuffer(TransportBuffer):
         fused: List[Tuple[torch.Tensor, int]] = []   # same-device: K1 writes
         copies: List[Tuple[int, int, int, int, int]] = []  # cross-device SDMA
         copies_2d: List[Tuple[int, int, int, int, int, int, int, int]] = []
+        # bounce coalescing: b_idx -> (staging tensor, pack pairs, used bytes)
+        bounce_state: Dict[int, Tuple[torch.Tensor, list, int]] = {}
         device = None
         for (kind, value), r, v in zip(self.payload, requests, values):
             if kind == "fetch_obj" or not isinstance(v, torch.Tensor):
@@ -412,6 +464,32 @@ class HipIpcTransportBuffer(TransportBuffer):
                 # windows already delivered during the handshake phases
                 chunks.release(value)
                 reply.append(("done", None))
+                continue
+            if kind == "bounce":
+                b_idx, off = value
+                bdesc = self.bounce_descs[b_idx]
+                device = v.device
+                nbytes = v.numel() * v.element_size()
+                if off + nbytes > bdesc.nbytes:
+                    raise RuntimeError(
+                        f"bounce overflow for {r.key}: {off}+{nbytes} > "
+                        f"{bdesc.nbytes}"
+                    )
+                entry = bounce_state.get(b_idx)
+                if entry is None:
+                    staging = torch.empty(
+                        bdesc.nbytes, dtype=torch.uint8, device=v.device
+                    )
+                    entry = (staging, [], 0)
+                staging, pairs, used = entry
+                slot = (
+                    staging[off : off + nbytes]
+                    .view(v.dtype)
+                    .reshape(v.shape)
+                )
+                pairs.append((v, slot))
+                bounce_state[b_idx] = (staging, pairs, max(used, off + nbytes))
+                reply.append(("bounced", None))
                 continue
             desc: IpcDescriptor = value
             device = v.device
@@ -463,12 +541,27 @@ class HipIpcTransportBuffer(TransportBuffer):
                      vc.numel() * vc.element_size())
                 )
                 self._hold.append(vc)
+        if bounce_state:
+            # pack ALL bounce pieces with one batched K1 launch, then move
+            # each bounce with ONE SDMA over xGMI into client memory
+            all_pairs = []
+            for staging, pairs, used in bounce_state.values():
+                all_pairs.extend(pairs)
+                self._hold.append(staging)
+            gpu_ops.copy_pairs(all_pairs, device, blocking=False)
+            for b_idx, (staging, _pairs, used) in bounce_state.items():
+                bdesc = self.bounce_descs[b_idx]
+                remote = cache.resolve(bdesc, device.index)
+                copies.append(
+                    (remote, bdesc.device_index, staging.data_ptr(),
+                     device.index, used)
+                )
         if copies or copies_2d:
             import asyncio
 
             if copies:
-                # K1 pack kernels ran on the current stream; the pool
-                # streams used by copy_batch must observe their writes
+                # pack kernels ran on the current stream; the pool streams
+                # used by copy_batch must observe their writes
                 torch.cuda.current_stream(device).synchronize()
                 await asyncio.to_thread(_run_copies, copies)
             if copies_2d:
@@ -487,6 +580,19 @@ class HipIpcTransportBuffer(TransportBuffer):
                 else:
                     out.append(value)
                 continue
+            if kind == "bounced":
+                b_idx, off = self.payload[i][1]
+                dest = r.tensor_val
+                nbytes = dest.numel() * dest.element_size()
+                piece = (
+                    self._bounces[b_idx][off : off + nbytes]
+                    .view(dest.dtype)
+                    .reshape(dest.shape)
+                )
+                scatter_pairs.append((piece, dest))
+                device = dest.device
+                out.append(dest)
+                continue
             scratch = self._scratch.get(i)
             if scratch is not None:
                 scatter_pairs.append((scratch, r.tensor_val))
@@ -502,4 +608,6 @@ class HipIpcTransportBuffer(TransportBuffer):
     async def drop(self) -> None:
         self._hold.clear()
         self._scratch.clear()
+        self._bounces = []
+        self.bounce_descs = []
         self.payload = None
