@@ -434,7 +434,7 @@ class HipIpcTransportBuffer(TransportBuffer):
                     self.bounce_descs.append(export_tensor(buf))
                     b_idx += 1
                     off = 0
-                payload[i] = ("bounce", (b_idx, off))
+                payload[i] = ("bounce", (b_idx, off, nbytes))
                 off += a
                 remaining -= a
             if device.index not in synced:
@@ -466,14 +466,14 @@ class HipIpcTransportBuffer(TransportBuffer):
                 reply.append(("done", None))
                 continue
             if kind == "bounce":
-                b_idx, off = value
+                b_idx, off, want_bytes = value
                 bdesc = self.bounce_descs[b_idx]
                 device = v.device
                 nbytes = v.numel() * v.element_size()
-                if off + nbytes > bdesc.nbytes:
+                if nbytes != want_bytes or off + nbytes > bdesc.nbytes:
                     raise RuntimeError(
-                        f"bounce overflow for {r.key}: {off}+{nbytes} > "
-                        f"{bdesc.nbytes}"
+                        f"bounce size mismatch for {r.key}: stored {nbytes} "
+                        f"vs reserved {want_bytes} at {off}/{bdesc.nbytes}"
                     )
                 entry = bounce_state.get(b_idx)
                 if entry is None:
@@ -581,7 +581,7 @@ class HipIpcTransportBuffer(TransportBuffer):
                     out.append(value)
                 continue
             if kind == "bounced":
-                b_idx, off = self.payload[i][1]
+                b_idx, off, _nb = self.payload[i][1]
                 dest = r.tensor_val
                 nbytes = dest.numel() * dest.element_size()
                 piece = (
