@@ -18,6 +18,7 @@ from __future__ import annotations
 
 import os
 import socket
+import time
 from dataclasses import dataclass
 from typing import Any, Dict, List, Optional, Sequence, Tuple, Union
 
@@ -240,8 +241,6 @@ class StorageVolume(Actor):
 
     @endpoint
     async def put(self, buffer: TransportBuffer, requests: Sequence[Request]):
-        import time
-
         t0 = time.perf_counter()
         buffer.attach_volume(self.ctx)
         existing = [self.store.find_existing(r) for r in requests]
@@ -258,8 +257,6 @@ class StorageVolume(Actor):
 
     @endpoint
     async def get(self, buffer: TransportBuffer, requests: Sequence[Request]):
-        import time
-
         t0 = time.perf_counter()
         buffer.attach_volume(self.ctx)
         values = [self.store.fetch(r) for r in requests]

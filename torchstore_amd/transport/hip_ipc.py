@@ -34,6 +34,7 @@ batch can be mixed — e.g. a state_dict with scalar stats).
 
 from __future__ import annotations
 
+import asyncio
 import os
 import uuid
 from dataclasses import dataclass
@@ -354,8 +355,6 @@ class HipIpcTransportBuffer(TransportBuffer):
         if copies:
             # executor thread: the volume keeps serving other clients while
             # the batched pull runs (the C++ side drops the GIL)
-            import asyncio
-
             await asyncio.to_thread(_run_copies, copies)
         return out
 
@@ -527,8 +526,6 @@ class HipIpcTransportBuffer(TransportBuffer):
                     self._hold.append(vc)
             reply.append(("done", None))
         if fused:
-            import asyncio
-
             # executor thread: loop stays responsive; kernel runs on the
             # device's default stream and is synchronized before return
             rejects = await asyncio.to_thread(
@@ -557,8 +554,6 @@ class HipIpcTransportBuffer(TransportBuffer):
                      device.index, used)
                 )
         if copies or copies_2d:
-            import asyncio
-
             if copies:
                 # pack kernels ran on the current stream; the pool streams
                 # used by copy_batch must observe their writes
