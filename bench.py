@@ -65,7 +65,19 @@ class GeneratorActor(Actor):
 
         v = self.dst_sd[name]
         local = v.tensor if isinstance(v, LocalShard) else v
-        return torch.equal(local, expected.to(local.device))
+        exp = expected.to(local.device)
+        ok = torch.equal(local, exp)
+        if not ok:
+            lf = local.float()
+            ef = exp.float()
+            print(
+                f"[generator verify {name}] got[:4]={lf.flatten()[:4].tolist()} "
+                f"exp[:4]={ef.flatten()[:4].tolist()} "
+                f"maxdiff={(lf - ef).abs().max().item()} "
+                f"nz_got={int((lf != 0).sum())}/{lf.numel()}",
+                flush=True,
+            )
+        return ok
 
 
 def parse_args():
