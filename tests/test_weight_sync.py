@@ -191,6 +191,48 @@ async def test_dtype_mismatch_raises(fake_codec):
     await _with_store(body)
 
 
+async def test_localshard_source_and_dest(fake_codec):
+    """The bench's mesh-free layout path: LocalShard entries on BOTH ends
+    (regression: these were silently skipped as non-Tensor leaves)."""
+    from torchstore_amd.types import LocalShard
+    from torchstore_amd.weight_sync import DirectWeightSyncSource
+
+    async def body():
+        c = ts.client()
+        full = torch.randn(8, 4)
+        sources = []
+        for r in range(2):
+            src = DirectWeightSyncSource(c, "sync", rank=r, world_size=2)
+            sd = {
+                "w": LocalShard(
+                    tensor=full[r * 4 : (r + 1) * 4].clone(),
+                    slice=TensorSlice(
+                        offsets=(r * 4, 0), local_shape=(4, 4),
+                        global_shape=(8, 4), coordinates=(r,), mesh_shape=(2,),
+                    ),
+                )
+            }
+            await src.push(sd)
+            sources.append(src)
+
+        dest = DirectWeightSyncDest(c, "sync")
+        # dest wants rows 2..6 — spans both source shards
+        dst = {
+            "w": LocalShard(
+                tensor=torch.zeros(4, 4),
+                slice=TensorSlice(
+                    offsets=(2, 0), local_shape=(4, 4), global_shape=(8, 4),
+                    coordinates=(0,), mesh_shape=(2,),
+                ),
+            )
+        }
+        await dest.pull(dst)
+        assert len(dest._plan) == 2
+        assert torch.equal(dst["w"].tensor, full[2:6])
+
+    await _with_store(body)
+
+
 async def test_missing_source_raises(fake_codec):
     async def body():
         c = ts.client()

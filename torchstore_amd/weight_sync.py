@@ -35,7 +35,7 @@ import torch
 from torchstore_amd.client import LocalClient
 from torchstore_amd.ops.slicing import byte_view, region_view
 from torchstore_amd.transport.hip_ipc import IpcDescriptor
-from torchstore_amd.types import Request, TensorSlice
+from torchstore_amd.types import LocalShard, Request, TensorSlice
 from torchstore_amd.utils.logging import LatencyTracker, get_logger
 
 logger = get_logger("torchstore_amd.weight_sync")
@@ -229,7 +229,7 @@ class DirectWeightSyncSource:
         codec = get_codec()
         handles: List[WeightHandle] = []
         for name, value in flat.items():
-            if not isinstance(value, torch.Tensor):
+            if not isinstance(value, (torch.Tensor, LocalShard)):
                 continue
             local, tslice = _request_slice(value)
             local = local.detach()
@@ -306,7 +306,7 @@ class DirectWeightSyncDest:
             by_name.setdefault(h.name, []).append(h)
         plan: List[_TransferOp] = []
         for name, value in dest_flat.items():
-            if not isinstance(value, torch.Tensor):
+            if not isinstance(value, (torch.Tensor, LocalShard)):
                 continue
             dest_local, dest_slice = _request_slice(value)
             wanted = dest_slice or _full_slice(dest_local.shape)
