@@ -60,24 +60,27 @@ class GeneratorActor(Actor):
         return time.perf_counter() - t0, nbytes
 
     @endpoint
-    def verify(self, name, expected):
+    def verify_pattern(self):
+        """Every pulled entry must equal the position-determined pattern —
+        verifies the cross-GPU read paths without exchanging data."""
+        from torchstore_amd.models import llama
         from torchstore_amd.types import LocalShard
 
-        v = self.dst_sd[name]
-        local = v.tensor if isinstance(v, LocalShard) else v
-        exp = expected.to(local.device)
-        ok = torch.equal(local, exp)
-        if not ok:
-            lf = local.float()
-            ef = exp.float()
-            print(
-                f"[generator verify {name}] got[:4]={lf.flatten()[:4].tolist()} "
-                f"exp[:4]={ef.flatten()[:4].tolist()} "
-                f"maxdiff={(lf - ef).abs().max().item()} "
-                f"nz_got={int((lf != 0).sum())}/{lf.numel()}",
-                flush=True,
+        bad = []
+        for name, v in self.dst_sd.items():
+            local = v.tensor if isinstance(v, LocalShard) else v
+            offsets = (
+                v.slice.offsets
+                if isinstance(v, LocalShard)
+                else (0,) * local.dim()
             )
-        return ok
+            exp = llama.expected_pattern(
+                tuple(local.shape), offsets, local.dtype, local.device
+            )
+            if not torch.equal(local, exp):
+                bad.append(name)
+        torch.cuda.synchronize()
+        return bad
 
 
 def parse_args():
@@ -147,7 +150,7 @@ async def run_bench(args, rank, world, local_rank, coord: Coord):
     layers = args.layers or llama.LAYERS
     src_sd = llama.make_local_shard_state_dict(
         rank, world, llama.fsdp_placement, device=device, layers=layers,
-        zero=False,
+        pattern=True,
     )
     shapes = llama.llama3_8b_shapes(layers)
     payload_bytes = llama.total_bytes(shapes, torch.bfloat16)
@@ -206,21 +209,33 @@ async def run_bench(args, rank, world, local_rank, coord: Coord):
         await one_step()
     barrier()
 
-    # correctness guard (outside the timed region): the norm weight has the
-    # same layout in both placements, so the pulled local shard must equal
-    # the pushed one bit-for-bit
-    probe = "model.norm.weight"
-
-    def _local(x):
-        return x.tensor if isinstance(x, LocalShard) else x
-
+    # correctness guard (outside the timed region): every pulled shard must
+    # equal the position-determined pattern — this validates the cross-GPU
+    # (xGMI) read paths end to end, not just co-located regions
     if args.mode == "direct":
-        ok = await gen.verify.call_one(probe, _local(src_sd[probe]).cpu())
-        if not ok:
-            raise RuntimeError("bench correctness probe failed (direct pull)")
+        bad = await gen.verify_pattern.call_one()
+        if bad:
+            raise RuntimeError(
+                f"bench correctness probe failed (direct pull): {bad[:5]}"
+            )
     else:
-        if not torch.equal(_local(dst_sd[probe]), _local(src_sd[probe])):
-            raise RuntimeError("bench correctness probe failed: pulled != pushed")
+        bad = []
+        for name, v in dst_sd.items():
+            local = v.tensor if isinstance(v, LocalShard) else v
+            offsets = (
+                v.slice.offsets if isinstance(v, LocalShard)
+                else (0,) * local.dim()
+            )
+            exp = llama.expected_pattern(
+                tuple(local.shape), offsets, local.dtype, local.device
+            )
+            if not torch.equal(local, exp):
+                bad.append(name)
+        if bad:
+            raise RuntimeError(
+                f"bench correctness probe failed (reshard pull): {bad[:5]} "
+                f"({len(bad)} of {len(dst_sd)} entries wrong)"
+            )
 
     t0 = time.perf_counter()
     for _ in range(args.steps):

@@ -69,6 +69,39 @@ def tp_placement(shape: Tuple[int, ...], world: int) -> Optional[int]:
     return None
 
 
+_PATTERN_COEFS = (7, 13, 29, 31)
+
+
+def pattern_fill(t: torch.Tensor, offsets) -> torch.Tensor:
+    """Fill a shard with values determined by GLOBAL position.
+
+    Any process can recompute any region independently, so cross-GPU
+    reshard paths can be verified without exchanging data:
+    ``v[i0,i1,..] = (((Σ coef_d * global_i_d) mod 61) - 30) * 0.01``.
+    """
+    idx = None
+    for d in range(t.dim()):
+        ar = (
+            torch.arange(
+                offsets[d], offsets[d] + t.shape[d],
+                device=t.device, dtype=torch.int64,
+            )
+            * _PATTERN_COEFS[d]
+        )
+        shape = [1] * t.dim()
+        shape[d] = -1
+        ar = ar.view(shape)
+        idx = ar if idx is None else idx + ar
+    vals = ((idx % 61) - 30).to(torch.float32) * 0.01
+    t.copy_(vals.to(t.dtype).expand_as(t))
+    return t
+
+
+def expected_pattern(shape, offsets, dtype, device) -> torch.Tensor:
+    out = torch.empty(shape, dtype=dtype, device=device)
+    return pattern_fill(out, offsets)
+
+
 def make_local_shard_state_dict(
     rank: int,
     world: int,
@@ -78,6 +111,7 @@ def make_local_shard_state_dict(
     layers: int = LAYERS,
     zero: bool = True,
     scale: int = 1,
+    pattern: bool = False,
 ):
     """DTensor-free sharded state_dict: {name: LocalShard | tensor}.
 
@@ -92,7 +126,9 @@ def make_local_shard_state_dict(
     for name, shape in shapes.items():
         if world == 1:
             t = torch.empty(shape, dtype=dtype, device=device)
-            if not zero:
+            if pattern:
+                pattern_fill(t, (0,) * len(shape))
+            elif not zero:
                 t.normal_(0, 0.02)
             out[name] = t
             continue
@@ -110,7 +146,9 @@ def make_local_shard_state_dict(
                 for d in range(len(shape))
             )
         t = torch.empty(local_shape, dtype=dtype, device=device)
-        if not zero:
+        if pattern:
+            pattern_fill(t, offsets)
+        elif not zero:
             t.normal_(0, 0.02)
         out[name] = LocalShard(
             tensor=t,
