@@ -72,13 +72,7 @@ def tp_placement(shape: Tuple[int, ...], world: int) -> Optional[int]:
 _PATTERN_COEFS = (7, 13, 29, 31)
 
 
-def pattern_fill(t: torch.Tensor, offsets) -> torch.Tensor:
-    """Fill a shard with values determined by GLOBAL position.
-
-    Any process can recompute any region independently, so cross-GPU
-    reshard paths can be verified without exchanging data:
-    ``v[i0,i1,..] = (((Σ coef_d * global_i_d) mod 61) - 30) * 0.01``.
-    """
+def _pattern_block(t: torch.Tensor, offsets) -> None:
     idx = None
     for d in range(t.dim()):
         ar = (
@@ -94,6 +88,29 @@ def pattern_fill(t: torch.Tensor, offsets) -> torch.Tensor:
         idx = ar if idx is None else idx + ar
     vals = ((idx % 61) - 30).to(torch.float32) * 0.01
     t.copy_(vals.to(t.dtype).expand_as(t))
+
+
+def pattern_fill(t: torch.Tensor, offsets) -> torch.Tensor:
+    """Fill a shard with values determined by GLOBAL position.
+
+    Any process can recompute any region independently, so cross-GPU
+    reshard paths can be verified without exchanging data:
+    ``v[i0,i1,..] = (((Σ coef_d * global_i_d) mod 61) - 30) * 0.01``.
+
+    Computed in row chunks: the int64 index grid of a big shard would
+    otherwise allocate a multi-GB transient whose cached allocator block
+    later hosts small tensors — pushing them over the ≥2 GiB IPC-mapping
+    limit and onto the slow windowed path.
+    """
+    max_elems = 16 << 20  # ≈128 MB of int64 transients per block
+    if t.dim() == 0 or t.numel() <= max_elems:
+        _pattern_block(t, tuple(offsets))
+        return t
+    row_elems = max(1, t[0].numel())
+    rows_per = max(1, max_elems // row_elems)
+    for r0 in range(0, t.shape[0], rows_per):
+        sub = t[r0 : r0 + rows_per]
+        _pattern_block(sub, (offsets[0] + r0,) + tuple(offsets[1:]))
     return t
 
 
