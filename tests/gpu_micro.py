@@ -69,6 +69,30 @@ def main():
         )
         print(f"  bw: {n/dt/1e9:.1f} GB/s")
 
+    # large-copy strategies: kernel vs SDMA variants
+    n1g = 1 << 30
+    a1 = torch.empty(n1g, dtype=torch.uint8, device="cuda")
+    b1 = torch.empty(n1g, dtype=torch.uint8, device="cuda")
+    for nsplit in [2, 4, 8, 16]:
+        chunk = n1g // nsplit
+        copies = [
+            (b1.data_ptr() + i * chunk, 0, a1.data_ptr() + i * chunk, 1, chunk)
+            for i in range(nsplit)
+        ]  # src_dev=1 forces the SDMA branch (ptrs are same-device; UVA ok)
+        dt = timeit(
+            f"sdma split x{nsplit} 1GiB", lambda: e.copy_batch(copies), n=10
+        )
+        print(f"  bw: {n1g/dt/1e9:.1f} GB/s")
+    dt = timeit(
+        "sdma 2d 1MBx1024 1GiB",
+        lambda: e.copy_batch_2d([
+            (b1.data_ptr(), 0, 1 << 20, a1.data_ptr(), 0, 1 << 20,
+             1 << 20, 1024)
+        ]),
+        n=10,
+    )
+    print(f"  bw: {n1g/dt/1e9:.1f} GB/s")
+
     # batched: 64 copies of 16MB
     srcs = [torch.empty(16 << 20, dtype=torch.uint8, device="cuda") for _ in range(64)]
     dsts = [torch.empty(16 << 20, dtype=torch.uint8, device="cuda") for _ in range(64)]
