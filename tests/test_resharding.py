@@ -104,6 +104,20 @@ async def test_fsdp_style_shard_replicate():
     await _reshard_case(4, (2, 2), ["0", "r"], 2, (2,), ["0"])
 
 
+async def test_uneven_shards_world3():
+    """16 rows over 3 ranks: chunks of 6/6/4 (torch chunk semantics) —
+    offsets and intersections must handle the ragged tail."""
+    await _reshard_case(3, (3,), ["0"], 2, (2,), ["0"], shape=(16, 16))
+
+
+async def test_uneven_shards_dim_change():
+    await _reshard_case(3, (3,), ["0"], 3, (3,), ["1"], shape=(16, 12))
+
+
+async def test_uneven_grow_world():
+    await _reshard_case(2, (2,), ["1"], 3, (3,), ["1"], shape=(8, 10))
+
+
 async def test_full_tensor_get_from_shards():
     """A rank outside any mesh fetches the assembled full tensor."""
     controller = await ts.initialize(

@@ -497,6 +497,12 @@ class HipIpcTransportBuffer(TransportBuffer):
                     f"get size mismatch for {r.key}: stored {tuple(v.shape)} "
                     f"vs dest {desc.shape}"
                 )
+            if v.dtype != desc.dtype:
+                raise TypeError(
+                    f"get dtype mismatch for {r.key}: stored {v.dtype} vs "
+                    f"dest {desc.dtype} — raw one-sided copies cannot cast; "
+                    "fetch at the stored dtype and cast locally"
+                )
             dst_ptr = cache.resolve(desc, v.device.index)
             if desc.device_index == v.device.index:
                 # co-located: the K1 gather kernel writes STRAIGHT into the
