@@ -285,12 +285,22 @@ class WeightSource(Actor):
     def mutate(self):
         with torch.no_grad():
             self.params["w"].add_(1.0)
+            self.params["master"].mul_(1.5)
         torch.cuda.synchronize()
         return "ok"
 
     @endpoint
     def snapshot(self):
         return {k: v.cpu() for k, v in self.params.items()}
+
+    @endpoint
+    async def push_cast(self):
+        """Separate sync key with an fp32→bf16 staging cast (K3)."""
+        await ts.put_state_dict(
+            {"master": self.params["master"]}, "dsync_cast", direct=True,
+            transfer_dtype=torch.bfloat16, rank=0, world_size=1,
+        )
+        return "ok"
 
 
 @requires_gpu
@@ -332,6 +342,25 @@ async def test_direct_weight_sync_real_ipc():
         snap2 = await src.snapshot.call_one()
         assert torch.equal(dst["w"].cpu(), snap2["w"])
         assert not torch.equal(snap["w"], snap2["w"])
+
+        # transfer-dtype staging: push casts fp32 master -> bf16 via K3;
+        # a re-push refreshes the staging after the in-place update
+        assert await src.push_cast.call_one() == "ok"
+        cast_dst = {"master": torch.zeros(256, 256, device="cuda",
+                                          dtype=torch.bfloat16)}
+        await ts.get_state_dict("dsync_cast", cast_dst, direct=True)
+        torch.cuda.synchronize()
+        assert torch.equal(
+            cast_dst["master"].cpu(), snap2["master"].to(torch.bfloat16)
+        )
+        await src.mutate.call_one()
+        assert await src.push_cast.call_one() == "ok"  # refresh re-casts
+        await ts.get_state_dict("dsync_cast", cast_dst, direct=True)
+        torch.cuda.synchronize()
+        snap3 = await src.snapshot.call_one()
+        assert torch.equal(
+            cast_dst["master"].cpu(), snap3["master"].to(torch.bfloat16)
+        )
     finally:
         if mesh is not None:
             await mesh.stop()
