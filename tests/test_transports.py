@@ -48,6 +48,42 @@ def test_rpc_buffer_carries_payload_out_of_band():
     assert torch.equal(back.data[0][1], t)
 
 
+def test_pin_fail_open(monkeypatch):
+    """hipHostRegister failures must warn once per code and keep copies
+    working unpinned (reference: _FakeCudart, test_shared_memory.py:342)."""
+    import warnings
+
+    from torchstore_amd.transport import shm
+
+    class FakeCudart:
+        def cudaHostRegister(self, ptr, nbytes, flags):
+            return 712  # hipErrorHostMemoryAlreadyRegistered-style failure
+
+    monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(torch.cuda, "cudart", lambda: FakeCudart())
+    monkeypatch.setattr(shm, "_PIN_WARNED", set())
+
+    seg = shm._allocate_segment(4096)
+    pinned = {}
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        shm._try_pin(seg, pinned)
+        shm._try_pin(seg, pinned)  # second call: same code, no new warning
+    assert pinned == {}  # fail-open: nothing recorded as pinned
+    msgs = [str(x.message) for x in w if "hipHostRegister" in str(x.message)]
+    assert len(msgs) == 1, msgs
+
+    # a succeeding register IS recorded
+    class GoodCudart:
+        def cudaHostRegister(self, ptr, nbytes, flags):
+            return 0
+
+    monkeypatch.setattr(torch.cuda, "cudart", lambda: GoodCudart())
+    shm._try_pin(seg, pinned)
+    assert len(pinned) == 1
+    pinned.clear()  # avoid unregister on a fake
+
+
 def test_context_strip_on_serialization():
     buf = RpcTransportBuffer()
     buf.bind_client(object(), TransportContext())
