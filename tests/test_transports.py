@@ -6,7 +6,7 @@ from torchstore_amd.runtime import serde
 from torchstore_amd.transport.hip_ipc import HipIpcTransportBuffer, IpcDescriptor
 from torchstore_amd.transport.shm import ShmDescriptor, ShmTransportBuffer
 from torchstore_amd.transport.rpc import RpcTransportBuffer
-from torchstore_amd.transport.base import TransportContext
+from torchstore_amd.transport.base import TransportContext, TransportType
 
 
 def test_ipc_buffer_strips_local_tensors():
@@ -90,3 +90,26 @@ def test_context_strip_on_serialization():
     header, bufs = serde.dumps(buf)
     back = serde.loads(header, bufs)
     assert back._client_ctx is None and back._volume_ref is None
+
+
+async def test_mutable_shm_zero_copy_get(monkeypatch):
+    """TORCHSTORE_AMD_MUTABLE_SHM=1: a get returns the live segment view —
+    a later put to the key is visible through the earlier result."""
+    monkeypatch.setenv("TORCHSTORE_AMD_MUTABLE_SHM", "1")
+    import torchstore_amd as ts
+    from torchstore_amd.strategy import SingletonStrategy
+
+    await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(transport=TransportType.SHARED_MEMORY),
+        storage_device="cpu",
+    )
+    try:
+        await ts.put("live", torch.ones(16))
+        view = await ts.get("live")
+        assert view.eq(1).all()
+        await ts.put("live", torch.full((16,), 5.0))
+        # the earlier result aliases the volume's segment: update visible
+        assert view.eq(5).all()
+    finally:
+        await ts.shutdown()

@@ -45,6 +45,14 @@ from torchstore_amd.utils.logging import get_logger
 logger = get_logger("torchstore_amd.shm")
 
 _PIN_WARNED: set = set()
+
+# opt-in zero-copy gets: return the LIVE typed view of the volume's segment
+# instead of a clone (the reference's TORCHSTORE_MUTABLE_SHM semantics —
+# later puts to the key become visible through previously returned tensors)
+def _mutable_shm() -> bool:
+    import os
+
+    return os.environ.get("TORCHSTORE_AMD_MUTABLE_SHM", "0") == "1"
 _HIP_HOST_REGISTER_PORTABLE = 1
 
 
@@ -403,7 +411,7 @@ class ShmTransportBuffer(TransportBuffer):
             typed = _typed_view(seg, desc)
             dest = r.tensor_val
             if dest is None:
-                out.append(typed.clone())
+                out.append(typed if _mutable_shm() else typed.clone())
                 continue
             if dest.device.type == "cuda":
                 _try_pin(seg, cache.pinned)
