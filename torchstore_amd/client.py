@@ -263,6 +263,7 @@ class LocalClient:
             sf.request.tensor_val = torch.empty(
                 meta.shape, dtype=meta.dtype, device=device
             )
+            sf.request.dest_owned = True
 
     @staticmethod
     def _alloc_device(meta: TensorMeta) -> torch.device:

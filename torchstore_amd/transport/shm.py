@@ -410,7 +410,7 @@ class ShmTransportBuffer(TransportBuffer):
             seg = cache.attach(desc)
             typed = _typed_view(seg, desc)
             dest = r.tensor_val
-            if dest is None:
+            if dest is None or (r.dest_owned and _mutable_shm()):
                 out.append(typed if _mutable_shm() else typed.clone())
                 continue
             if dest.device.type == "cuda":

@@ -132,6 +132,9 @@ class Request:
     is_object: bool = False
     # hint: land the fetch directly in tensor_val (in-place get)
     inplace: bool = False
+    # tensor_val was allocated by the planner (not user memory) — transports
+    # may substitute a zero-copy result for it
+    dest_owned: bool = False
 
     @classmethod
     def from_any(cls, key: str, value: Any) -> "Request":
