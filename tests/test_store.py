@@ -197,3 +197,41 @@ async def test_large_tensor_roundtrip(transport):
         assert torch.equal(out, t)
 
     await _with_store(transport, body)
+
+
+@pytest.mark.parametrize(
+    "strategy_name,num_volumes",
+    [("singleton", 1), ("local_rank", 2), ("host", 1)],
+)
+@pytest.mark.parametrize(
+    "matrix_transport", [TransportType.RPC, TransportType.SHARED_MEMORY]
+)
+async def test_transport_strategy_matrix(strategy_name, num_volumes, matrix_transport):
+    """The reference's transport x strategy product (tests/utils.py:63-69):
+    core put/get/batch semantics must hold for every combination."""
+    from torchstore_amd.strategy import (
+        HostStrategy,
+        LocalRankStrategy,
+        SingletonStrategy,
+    )
+
+    strategy = {
+        "singleton": SingletonStrategy,
+        "local_rank": LocalRankStrategy,
+        "host": HostStrategy,
+    }[strategy_name](transport=matrix_transport)
+    await ts.initialize(
+        num_storage_volumes=num_volumes, strategy=strategy, storage_device="cpu"
+    )
+    try:
+        t = torch.randn(32, 32)
+        await ts.put("m/x", t)
+        assert torch.equal(await ts.get("m/x"), t)
+        await ts.put_batch({"m/y": torch.ones(4), "m/obj": {"a": 1}})
+        out = await ts.get_batch({"m/y": None, "m/obj": None})
+        assert out["m/y"].eq(1).all() and out["m/obj"] == {"a": 1}
+        assert sorted(await ts.keys("m")) == ["m/obj", "m/x", "m/y"]
+        await ts.delete("m/x")
+        assert not await ts.exists("m/x")
+    finally:
+        await ts.shutdown()
