@@ -381,6 +381,12 @@ class DirectWeightSyncDest:
                 )
         return plan
 
+    def invalidate(self) -> None:
+        """Drop the cached plan + handles (call when the source re-registers,
+        e.g. after a trainer restart)."""
+        self._plan = None
+        self._handles = None
+
     async def pull(self, dest_state_dict: Dict[str, Any]) -> None:
         codec = get_codec()
         if self._plan is None:
