@@ -89,3 +89,27 @@ async def test_double_shutdown_is_safe():
 async def test_client_before_initialize_raises():
     with pytest.raises(RuntimeError, match="not initialized"):
         ts.client("never-made")
+
+
+async def test_two_named_stores_isolated():
+    """Two stores in one process: key spaces and lifecycles are independent."""
+    await ts.initialize(
+        num_storage_volumes=1, strategy=SingletonStrategy(),
+        storage_device="cpu", store_name="alpha",
+    )
+    await ts.initialize(
+        num_storage_volumes=1, strategy=SingletonStrategy(),
+        storage_device="cpu", store_name="beta",
+    )
+    try:
+        await ts.put("k", torch.ones(4), store_name="alpha")
+        await ts.put("k", torch.full((4,), 2.0), store_name="beta")
+        assert (await ts.get("k", store_name="alpha")).eq(1).all()
+        assert (await ts.get("k", store_name="beta")).eq(2).all()
+        assert not await ts.exists("other", store_name="alpha")
+        await ts.shutdown("alpha")
+        # beta unaffected by alpha's teardown
+        assert (await ts.get("k", store_name="beta")).eq(2).all()
+    finally:
+        await ts.shutdown("beta")
+        await ts.shutdown("alpha")  # idempotent
