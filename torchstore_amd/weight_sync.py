@@ -279,6 +279,7 @@ class DirectWeightSyncDest:
         self.key = key
         self._plan: Optional[List[_TransferOp]] = None
         self._handles: Optional[List[WeightHandle]] = None
+        self._plan_dest_id: Optional[int] = None
 
     async def _fetch_handles(self) -> List[WeightHandle]:
         if self._handles is not None:
@@ -389,10 +390,15 @@ class DirectWeightSyncDest:
 
     async def pull(self, dest_state_dict: Dict[str, Any]) -> None:
         codec = get_codec()
+        if self._plan_dest_id is not None and self._plan_dest_id != id(dest_state_dict):
+            # a different destination dict: the cached plan points at the old
+            # tensors — rebuild against the new ones
+            self._plan = None
         if self._plan is None:
             handles = await self._fetch_handles()
             dest_flat, _ = _flatten(dest_state_dict)
             self._plan = self._build_plan(handles, dest_flat)
+            self._plan_dest_id = id(dest_state_dict)
             logger.info(
                 "direct sync plan: %d ops (%d zero-copy, %d pitched, %d recv)",
                 len(self._plan),
