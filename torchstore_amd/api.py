@@ -71,6 +71,9 @@ async def initialize(
             f"{store_name}-volume",
             volume_id_seed=strategy.volume_id_seed,
             device=storage_device,
+            # a node bringing up many ranks + volumes at once can take a
+            # while per torch import; don't flake on a loaded box
+            timeout=240.0,
         ),
         asyncio.to_thread(
             spawn_actor, Controller, f"{store_name}-controller", store_name

@@ -109,7 +109,7 @@ async def initialize_spmd(
     env: Optional[SPMDEnv] = None,
     store_name: str = api.DEFAULT_STORE,
     storage_device: str = "auto",
-    timeout_s: float = 120.0,
+    timeout_s: float = 300.0,
 ) -> ActorHandle:
     """Collective store bring-up across a torchrun world.
 
