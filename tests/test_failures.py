@@ -113,3 +113,22 @@ async def test_two_named_stores_isolated():
     finally:
         await ts.shutdown("beta")
         await ts.shutdown("alpha")  # idempotent
+
+
+@pytest.mark.slow
+async def test_repeated_lifecycle_no_leaks():
+    """Repeated bring-up/teardown in one process must not leak processes,
+    connections, or store state."""
+    import multiprocessing
+
+    for i in range(4):
+        await ts.initialize(
+            num_storage_volumes=1, strategy=SingletonStrategy(),
+            storage_device="cpu",
+        )
+        await ts.put("cycle", torch.full((8,), float(i)))
+        assert (await ts.get("cycle")).eq(float(i)).all()
+        await ts.shutdown()
+    # no stray actor children left behind
+    leftovers = [p for p in multiprocessing.active_children()]
+    assert leftovers == [], leftovers
