@@ -50,6 +50,10 @@ async def _write_frame(
     req_id: int,
     obj: Any,
 ) -> None:
+    # bufs alias live tensor memory (zero-copy): the memoryviews keep the
+    # storage alive until flushed; a concurrent in-place overwrite of a
+    # stored tensor during a reply can tear VALUES (never memory safety) —
+    # the same window the reference documents for mutable SHM reads
     header, bufs = serde.dumps(obj)
     async with lock:
         writer.write(_HDR.pack(kind, req_id, len(bufs), len(header)))
