@@ -269,7 +269,7 @@ async def run_bench(args, rank, world, local_rank, coord: Coord):
             "scaling": "strong",
             "vs_baseline": None,
             "dtype": "bf16",
-            "data": "synthetic (random-init Llama-3-8B shapes)",
+            "data": "synthetic (Llama-3-8B shapes, position-pattern values so every rank verifies its resharded pull bit-exactly)",
             "config": {
                 "model": "Llama-3-8B",
                 "global_batch": None,
