@@ -211,20 +211,29 @@ class HipIpcTransportBuffer(TransportBuffer):
         staging_desc = await volume.handshake.call_one(
             self, (token, tuple(t.shape), t.dtype), "chunk_put_init"
         )
-        staging_ptr = cache.resolve(staging_desc, t.device.index)
-        nbytes = t.numel() * t.element_size()
-        base = t.data_ptr()
-        off = 0
-        while off < nbytes:
-            win = min(CHUNK_BYTES, nbytes - off)
-            _run_copies(
-                [(staging_ptr, staging_desc.device_index,
-                  base + off, t.device.index, win)]
-            )
-            await volume.handshake.call_one(
-                self, (token, off, win), "chunk_put_commit"
-            )
-            off += win
+        try:
+            staging_ptr = cache.resolve(staging_desc, t.device.index)
+            nbytes = t.numel() * t.element_size()
+            base = t.data_ptr()
+            off = 0
+            while off < nbytes:
+                win = min(CHUNK_BYTES, nbytes - off)
+                _run_copies(
+                    [(staging_ptr, staging_desc.device_index,
+                      base + off, t.device.index, win)]
+                )
+                await volume.handshake.call_one(
+                    self, (token, off, win), "chunk_put_commit"
+                )
+                off += win
+        except BaseException:
+            # abort: return the staging chunk to the pool (uniflow's abort
+            # phase — a failed transfer must not leak volume resources)
+            try:
+                await volume.handshake.call_one(self, token, "chunk_release")
+            except Exception:  # noqa: BLE001
+                pass
+            raise
         return token
 
     async def _chunked_get_windows(
@@ -236,20 +245,27 @@ class HipIpcTransportBuffer(TransportBuffer):
         staging_desc = await volume.handshake.call_one(
             self, (token, request.meta_only()), "chunk_get_init"
         )
-        staging_ptr = cache.resolve(staging_desc, dest.device.index)
-        nbytes = dest.numel() * dest.element_size()
-        base = dest.data_ptr()
-        off = 0
-        while off < nbytes:
-            win = min(CHUNK_BYTES, nbytes - off)
-            await volume.handshake.call_one(
-                self, (token, off, win), "chunk_get_fill"
-            )
-            _run_copies(
-                [(base + off, dest.device.index,
-                  staging_ptr, staging_desc.device_index, win)]
-            )
-            off += win
+        try:
+            staging_ptr = cache.resolve(staging_desc, dest.device.index)
+            nbytes = dest.numel() * dest.element_size()
+            base = dest.data_ptr()
+            off = 0
+            while off < nbytes:
+                win = min(CHUNK_BYTES, nbytes - off)
+                await volume.handshake.call_one(
+                    self, (token, off, win), "chunk_get_fill"
+                )
+                _run_copies(
+                    [(base + off, dest.device.index,
+                      staging_ptr, staging_desc.device_index, win)]
+                )
+                off += win
+        except BaseException:
+            try:
+                await volume.handshake.call_one(self, token, "chunk_release")
+            except Exception:  # noqa: BLE001
+                pass
+            raise
         return token
 
     # -- volume handshake dispatcher --------------------------------------
