@@ -13,7 +13,7 @@ import torch
 from torchstore_amd.utils.net import pick_free_port
 
 
-def _spmd_worker(rank, world, master_port, result_path):
+def _spmd_worker(rank, world, master_port, result_path, strategy_name="local_rank"):
     os.environ.update(
         {
             "RANK": str(rank),
@@ -26,8 +26,13 @@ def _spmd_worker(rank, world, master_port, result_path):
     )
     import torchstore_amd as ts
 
+    strategy = {
+        "local_rank": ts.LocalRankStrategy,
+        "host": ts.HostStrategy,
+    }[strategy_name]()
+
     async def main():
-        await ts.initialize_spmd()
+        await ts.initialize_spmd(strategy=strategy)
         t = torch.full((16,), float(rank))
         await ts.put(f"rank{rank}/data", t)
         # cross-rank visibility: wait for the peer's key
@@ -52,7 +57,8 @@ def _spmd_worker(rank, world, master_port, result_path):
         json.dump(result, f)
 
 
-def test_spmd_lifecycle():
+@pytest.mark.parametrize("strategy_name", ["local_rank", "host"])
+def test_spmd_lifecycle(strategy_name):
     world = 2
     port = pick_free_port()
     ctx = mp.get_context("spawn")
@@ -62,7 +68,8 @@ def test_spmd_lifecycle():
         path = tempfile.mktemp(prefix=f"spmd-res-{rank}")
         paths.append(path)
         p = ctx.Process(
-            target=_spmd_worker, args=(rank, world, port, path)
+            target=_spmd_worker,
+            args=(rank, world, port, path, strategy_name),
         )
         p.start()
         procs.append(p)
