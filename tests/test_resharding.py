@@ -176,3 +176,32 @@ async def test_partial_commit_blocks_get():
             await put_mesh.stop()
         await ts.shutdown()
         await close_connections()
+
+
+async def test_repush_with_new_mesh_replaces_old_epoch():
+    """Re-pushing a key under a DIFFERENT mesh must fully replace the old
+    sharding — stale old-layout shards must never serve reads."""
+    controller = await ts.initialize(
+        num_storage_volumes=4,
+        strategy=LocalRankStrategy(),
+        storage_device="cpu",
+    )
+    m2 = m4 = None
+    try:
+        m2 = await _spawn_world(2, controller, "ep2")
+        m4 = await _spawn_world(4, controller, "ep4")
+        res = await m2.put_dtensor.call("w", (16, 16), (2,), ["0"], False, 1.0)
+        assert all(r == "ok" for r in res)
+        # new epoch: same key, 4-way mesh, DIFFERENT values
+        res = await m4.put_dtensor.call("w", (16, 16), (4,), ["1"], False, 2.0)
+        assert all(r == "ok" for r in res)
+        from tests.utils import make_full_tensor
+
+        out = await ts.get("w")
+        assert torch.equal(out, make_full_tensor((16, 16)) * 2.0)
+    finally:
+        for m in (m2, m4):
+            if m is not None:
+                await m.stop()
+        await ts.shutdown()
+        await close_connections()

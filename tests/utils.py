@@ -78,8 +78,9 @@ class DTensorWorker(Actor):
         mesh_shape: Tuple[int, ...],
         placements: Sequence[str],
         skip: bool = False,
+        value_scale: float = 1.0,
     ):
-        full = make_full_tensor(shape)
+        full = make_full_tensor(shape) * value_scale
         dt = self._dtensor(full, mesh_shape, placements)
         if skip:
             return "skipped"

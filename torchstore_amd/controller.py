@@ -43,8 +43,18 @@ class StorageInfo:
             # last write wins on type change (e.g. object overwritten by tensor)
             self.object_type = other.object_type
             self.tensor_slices = set(other.tensor_slices)
-        else:
-            self.tensor_slices |= other.tensor_slices
+            return
+        if other.tensor_slices and self.tensor_slices:
+            mine = next(iter(self.tensor_slices))
+            new = next(iter(other.tensor_slices))
+            if (
+                mine.mesh_shape != new.mesh_shape
+                or mine.global_shape != new.global_shape
+            ):
+                # new sharding epoch replaces the old layout
+                self.tensor_slices = set(other.tensor_slices)
+                return
+        self.tensor_slices |= other.tensor_slices
 
 
 @dataclass
@@ -120,6 +130,23 @@ class Controller(Actor):
             if locations is None:
                 locations = {}
                 self.index[r.key] = locations
+            if info.object_type == ObjectType.TENSOR_SLICE and locations:
+                new_slice = next(iter(info.tensor_slices))
+                stale = [
+                    vid for vid, old in locations.items()
+                    if old.object_type == ObjectType.TENSOR_SLICE
+                    and old.tensor_slices
+                    and (
+                        next(iter(old.tensor_slices)).mesh_shape
+                        != new_slice.mesh_shape
+                        or next(iter(old.tensor_slices)).global_shape
+                        != new_slice.global_shape
+                    )
+                ]
+                # a different mesh/global shape starts a new sharding epoch:
+                # drop every volume's old-layout entries for the key
+                for vid in stale:
+                    del locations[vid]
             if volume_id in locations:
                 locations[volume_id].merge(info)
             else:

@@ -115,6 +115,15 @@ class InMemoryStore(StorageImpl):
             entry = _ShardEntry()
             self.kv[key] = entry
         ts = request.tensor_slice
+        if entry.shards:
+            any_slice, _t = next(iter(entry.shards.values()))
+            if (
+                any_slice.mesh_shape != ts.mesh_shape
+                or any_slice.global_shape != ts.global_shape
+            ):
+                # a re-push under a different sharding: stale shards of the
+                # old layout could otherwise serve wrong regions
+                entry.shards.clear()
         entry.shards[ts.coordinates] = (ts, tensor)
 
     def find_existing(self, request: Request) -> Optional[torch.Tensor]:
