@@ -39,7 +39,7 @@ from torchstore_amd.strategy import (
     SingletonStrategy,
 )
 from torchstore_amd.transport import TransportType
-from torchstore_amd.types import Request, TensorSlice
+from torchstore_amd.types import LocalShard, Request, TensorSlice
 from torchstore_amd.utils.logging import init_logging
 
 init_logging()
@@ -70,6 +70,7 @@ __all__ = [
     "PlacementStrategy",
     "SingletonStrategy",
     "TransportType",
+    "LocalShard",
     "Request",
     "TensorSlice",
 ]
