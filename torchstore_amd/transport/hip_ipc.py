@@ -384,6 +384,29 @@ class HipIpcTransportBuffer(TransportBuffer):
                 return 0
         return -1
 
+    async def _stage_get_normal(
+        self, i: int, r: Request, synced: set
+    ) -> Tuple[str, Any]:
+        """Per-request direct staging: export the dest (or a dense scratch),
+        falling back to the windowed path for >=2 GiB blocks."""
+        dest = r.tensor_val
+        if dest.is_contiguous():
+            target = dest
+        else:
+            target = torch.empty(dest.shape, dtype=dest.dtype, device=dest.device)
+            self._scratch[i] = target
+        if target.device.index not in synced:
+            # pending client kernels touching dest must finish before the
+            # volume's one-sided writes land in it
+            torch.cuda.current_stream(target.device).synchronize()
+            synced.add(target.device.index)
+        self._hold.append(target)
+        desc = try_export(target)
+        if desc is None:
+            token = await self._chunked_get_windows(r, target)
+            return ("chunked", token)
+        return ("ipc", desc)
+
     async def client_stage_get(self, requests: Sequence[Request]) -> None:
         payload: List[Tuple[str, Any]] = []
         synced: set = set()
@@ -410,25 +433,7 @@ class HipIpcTransportBuffer(TransportBuffer):
                 payload.append(("bounce", None))  # offsets assigned below
                 bounce_plan.append((i, nbytes))
                 continue
-            if dest.is_contiguous():
-                target = dest
-            else:
-                target = torch.empty(
-                    dest.shape, dtype=dest.dtype, device=dest.device
-                )
-                self._scratch[i] = target
-            if target.device.index not in synced:
-                # pending client kernels touching dest must finish before the
-                # volume's one-sided writes land in it
-                torch.cuda.current_stream(target.device).synchronize()
-                synced.add(target.device.index)
-            self._hold.append(target)
-            desc = try_export(target)
-            if desc is None:
-                token = await self._chunked_get_windows(r, target)
-                payload.append(("chunked", token))
-            else:
-                payload.append(("ipc", desc))
+            payload.append(await self._stage_get_normal(i, r, synced))
 
         self.bounce_descs: List[IpcDescriptor] = []
         self._bounces: List[torch.Tensor] = []
@@ -440,19 +445,33 @@ class HipIpcTransportBuffer(TransportBuffer):
             off = 0
             size = 0
             b_idx = -1
+            ok = True
             for i, nbytes, a in aligned:
                 if b_idx < 0 or off + a > size:
                     size = min(cap, max(remaining, a))
                     buf = torch.empty(size, dtype=torch.uint8, device=device)
+                    desc_b = try_export(buf)
+                    if desc_b is None:
+                        # the bounce itself landed in a >=2 GiB cached block:
+                        # degrade to per-request staging instead of failing
+                        ok = False
+                        break
                     self._bounces.append(buf)
                     self._hold.append(buf)
-                    self.bounce_descs.append(export_tensor(buf))
+                    self.bounce_descs.append(desc_b)
                     b_idx += 1
                     off = 0
                 payload[i] = ("bounce", (b_idx, off, nbytes))
                 off += a
                 remaining -= a
-            if device.index not in synced:
+            if not ok:
+                self._bounces = []
+                self.bounce_descs = []
+                for i, nbytes, a in aligned:
+                    payload[i] = await self._stage_get_normal(
+                        i, requests[i], synced
+                    )
+            elif device.index not in synced:
                 torch.cuda.current_stream(device).synchronize()
         self.payload = payload
 
