@@ -104,6 +104,49 @@ def bounding_box(parts: Sequence[Region]) -> Region:
     return tuple(lo), tuple(hi[d] - lo[d] for d in range(ndim))
 
 
+def regions_disjoint(regions: Sequence[Region]) -> bool:
+    """Pairwise-disjointness of N-d regions (O(k²), k small in practice)."""
+    for i in range(len(regions)):
+        for j in range(i + 1, len(regions)):
+            if overlap(regions[i][0], regions[i][1],
+                       regions[j][0], regions[j][1]) is not None:
+                return False
+    return True
+
+
+def union_volume(regions: Sequence[Region]) -> int:
+    """Exact element count of the union of N-d regions.
+
+    Coordinate compression: each dim's boundaries split space into at most
+    2k-1 intervals; a grid cell is covered iff any region contains it.
+    Cost O((2k)^ndim · k) — fine for the shard counts resharding produces.
+    """
+    import itertools
+
+    ndim = len(regions[0][0])
+    bounds = [
+        sorted(
+            {r[0][d] for r in regions} | {r[0][d] + r[1][d] for r in regions}
+        )
+        for d in range(ndim)
+    ]
+    cells = [list(zip(b[:-1], b[1:])) for b in bounds]
+    total = 0
+    for cell in itertools.product(*cells):
+        if any(
+            all(
+                r[0][d] <= cell[d][0] and cell[d][1] <= r[0][d] + r[1][d]
+                for d in range(ndim)
+            )
+            for r in regions
+        ):
+            vol = 1
+            for lo, hi in cell:
+                vol *= hi - lo
+            total += vol
+    return total
+
+
 def assemble(
     parts: Sequence[Tuple[Tuple[int, ...], torch.Tensor]],
     out: Optional[torch.Tensor] = None,
@@ -128,7 +171,14 @@ def assemble(
         total = 1
         for s in shape:
             total *= s
-        covered = sum(t.numel() for _, t in parts)
+        # disjoint parts (the usual case): numel sum is the exact coverage.
+        # Partially overlapping parts could hide a gap behind the sum, so
+        # they get the exact (compressed-grid) union volume instead —
+        # a gap would otherwise return uninitialised torch.empty memory.
+        if regions_disjoint(regions):
+            covered = sum(t.numel() for _, t in parts)
+        else:
+            covered = union_volume(regions)
         if covered < total:
             raise ValueError(
                 f"parts cover {covered} elements but bounding box has {total}: "

@@ -310,7 +310,8 @@ class HipIpcTransportBuffer(TransportBuffer):
     # ------------------------------------------------------------- put --
     async def client_stage_put(self, requests: Sequence[Request]) -> None:
         payload: List[Tuple[str, Any]] = []
-        synced: set = set()
+        staged: List[Tuple[int, torch.Tensor]] = []
+        devices: set = set()
         for r in requests:
             if r.is_object:
                 payload.append(("inline", r.objects))
@@ -321,16 +322,22 @@ class HipIpcTransportBuffer(TransportBuffer):
                 continue
             tc = t.contiguous()
             self._hold.append(tc)
-            if t.device.index not in synced:
-                # writes producing t must be visible before the volume pulls
-                torch.cuda.current_stream(t.device).synchronize()
-                synced.add(t.device.index)
+            devices.add(tc.device.index)
+            payload.append(("pending", None))
+            staged.append((len(payload) - 1, tc))
+        # every producing kernel AND every pack copy from .contiguous() must
+        # be visible before the volume's one-sided pulls read the staging
+        # memory from another process — so the sync happens once per device
+        # AFTER the whole staging loop, not at the first tensor seen
+        for di in devices:
+            torch.cuda.current_stream(torch.device("cuda", di)).synchronize()
+        for i, tc in staged:
             desc = try_export(tc)
             if desc is None:
                 token = await self._chunked_put_windows(tc)
-                payload.append(("chunked", token))
+                payload[i] = ("chunked", token)
             else:
-                payload.append(("ipc", desc))
+                payload[i] = ("ipc", desc)
         self.payload = payload
 
     async def volume_receive(self, requests, existing, device):
@@ -441,6 +448,12 @@ class HipIpcTransportBuffer(TransportBuffer):
             device = requests[bounce_plan[0][0]].tensor_val.device
             cap = 1 << 30  # each bounce buffer stays well under the 2GiB limit
             aligned = [(i, n, (n + 255) & ~255) for i, n in bounce_plan]
+            # a single piece larger than the cap can never fit a bounce
+            # reservation — stage it directly (scratch + windowed path)
+            oversized = [e for e in aligned if e[2] > cap]
+            aligned = [e for e in aligned if e[2] <= cap]
+            for i, _n, _a in oversized:
+                payload[i] = await self._stage_get_normal(i, requests[i], synced)
             remaining = sum(a for _, _, a in aligned)
             off = 0
             size = 0

@@ -63,19 +63,36 @@ class PairInfo:
     backend: str
 
 
+@dataclass
+class PairEntry:
+    info: PairInfo
+    store: Any                    # TCPStore master
+    pg: Any = None                # created on volume arrival
+    confirmed: bool = False       # promoted only after a data op succeeds
+    inflight: int = 0
+    next_tag: int = 0
+    op_lock: Any = None           # asyncio.Lock — serializes RCCL ops
+
+
 class PgClientCache(TransportCache):
-    """Client side: hosts the TCPStore master + rank-0 PG per volume."""
+    """Client side: hosts the TCPStore master + rank-0 PG per volume.
+
+    Uniflow-protocol semantics (reference
+    ``transport/torchcomms/uniflow_buffer.py:88-116``): a freshly created
+    pair stays *unconfirmed* until the first data operation over it
+    succeeds; a failed rendezvous/op discards the unconfirmed entry so it
+    can never poison later reuse.
+    """
 
     def __init__(self):
         import threading
 
-        # volume_id -> (PairInfo, TCPStore, pg)
-        self.pairs: Dict[str, Tuple[PairInfo, Any, Any]] = {}
+        self.pairs: Dict[str, PairEntry] = {}
         self.lock = threading.Lock()
 
-    def get_or_create(self, volume_id: str, backend: str):
+    def get_or_create(self, volume_id: str, backend: str) -> PairEntry:
         entry = self.pairs.get(volume_id)
-        if entry is not None and entry[0].backend == backend:
+        if entry is not None and entry.info.backend == backend:
             return entry
         host = os.environ.get("TORCHSTORE_AMD_PG_HOST") or _local_addr()
         port = pick_free_port(host)
@@ -86,18 +103,28 @@ class PgClientCache(TransportCache):
         info = PairInfo(
             pair_id=uuid.uuid4().hex, host=host, port=port, backend=backend
         )
-        entry = (info, store, None)  # pg created on volume arrival
+        entry = PairEntry(info=info, store=store, op_lock=asyncio.Lock())
         self.pairs[volume_id] = entry
         return entry
 
     def ensure_pg(self, volume_id: str):
         """BLOCKS until the volume's rank-1 joins — call from an executor."""
         with self.lock:
-            info, store, pg = self.pairs[volume_id]
-            if pg is None:
-                pg = _make_pg(info.backend, store, rank=0)
-                self.pairs[volume_id] = (info, store, pg)
-            return pg
+            entry = self.pairs[volume_id]
+            if entry.pg is None:
+                entry.pg = _make_pg(entry.info.backend, entry.store, rank=0)
+            return entry.pg
+
+    def on_op_failed(self, volume_id: str, pair_id: str) -> None:
+        """Discard an unconfirmed pair a failed op created (publish-on-success)."""
+        entry = self.pairs.get(volume_id)
+        if (
+            entry is not None
+            and entry.info.pair_id == pair_id
+            and not entry.confirmed
+            and entry.inflight == 0
+        ):
+            del self.pairs[volume_id]
 
     def drop_key(self, key: str) -> None:
         return None
@@ -140,20 +167,32 @@ def _local_addr() -> str:
         return "127.0.0.1"
 
 
-async def _send_all(pg, tensors: List[torch.Tensor], dst: int) -> None:
+async def _send_all(
+    pg, tensors: List[torch.Tensor], dst: int, tag: int
+) -> None:
     def run():
         for t in tensors:
-            pg.send([t], dst, 0).wait()
+            pg.send([t], dst, tag).wait()
 
     await asyncio.get_running_loop().run_in_executor(None, run)
 
 
-async def _recv_all(pg, tensors: List[torch.Tensor], src: int) -> None:
+async def _recv_all(
+    pg, tensors: List[torch.Tensor], src: int, tag: int
+) -> None:
     def run():
         for t in tensors:
-            pg.recv([t], src, 0).wait()
+            pg.recv([t], src, tag).wait()
 
     await asyncio.get_running_loop().run_in_executor(None, run)
+
+
+class _NullLock:
+    async def __aenter__(self):
+        return self
+
+    async def __aexit__(self, *exc):
+        return False
 
 
 class _PgTransportBuffer(TransportBuffer):
@@ -162,6 +201,11 @@ class _PgTransportBuffer(TransportBuffer):
     def __init__(self):
         super().__init__()
         self.pair_info: Optional[PairInfo] = None
+        # per-operation tag: several sub-batches of one state_dict op run
+        # concurrently over the SAME cached pair; gloo matches send/recv
+        # by tag so cross-op pairing is unambiguous.  RCCL ignores tags —
+        # there the per-pair op_lock serializes whole operations instead.
+        self.op_tag: int = 0
         # aligned with requests: ("pg", meta) | ("inline", value)
         self.payload: Optional[List[Tuple[str, Any]]] = None
         self._send_task: Optional[asyncio.Task] = None
@@ -171,14 +215,32 @@ class _PgTransportBuffer(TransportBuffer):
             return t.contiguous().cpu()
         return t.contiguous()
 
+    def _begin_op(self) -> PairEntry:
+        cache: PgClientCache = self._client_ctx.cache(PgClientCache)
+        entry = cache.get_or_create(self._volume_ref.volume_id, self.backend)
+        self.pair_info = entry.info
+        self.op_tag = entry.next_tag
+        entry.next_tag = (entry.next_tag + 1) % (1 << 30)
+        entry.inflight += 1
+        return entry
+
+    def _op_serializer(self, entry: PairEntry):
+        # RCCL send/recv carry no tags: one op at a time per pair
+        return entry.op_lock if self.backend == "nccl" else _NullLock()
+
+    def _end_op(self, entry: PairEntry, ok: bool) -> None:
+        cache: PgClientCache = self._client_ctx.cache(PgClientCache)
+        entry.inflight -= 1
+        if ok:
+            entry.confirmed = True
+        else:
+            cache.on_op_failed(self._volume_ref.volume_id, entry.info.pair_id)
+
     # ------------------------------------------------------------- put --
     async def put(self, requests: Sequence[Request]) -> None:
         volume = self._volume_ref.volume
         cache: PgClientCache = self._client_ctx.cache(PgClientCache)
-        info, _store, _pg = cache.get_or_create(
-            self._volume_ref.volume_id, self.backend
-        )
-        self.pair_info = info
+        entry = self._begin_op()
         payload: List[Tuple[str, Any]] = []
         to_send: List[torch.Tensor] = []
         for r in requests:
@@ -196,14 +258,22 @@ class _PgTransportBuffer(TransportBuffer):
             pg = await asyncio.get_running_loop().run_in_executor(
                 None, cache.ensure_pg, self._volume_ref.volume_id
             )
-            await _send_all(pg, to_send, dst=1)
+            await _send_all(pg, to_send, dst=1, tag=self.op_tag)
 
-        send_task = asyncio.create_task(sender()) if to_send else None
+        ok = False
         try:
-            await volume.put.call_one(self, [r.meta_only() for r in requests])
+            async with self._op_serializer(entry):
+                send_task = asyncio.create_task(sender()) if to_send else None
+                try:
+                    await volume.put.call_one(
+                        self, [r.meta_only() for r in requests]
+                    )
+                finally:
+                    if send_task is not None:
+                        await send_task
+            ok = True
         finally:
-            if send_task is not None:
-                await send_task
+            self._end_op(entry, ok)
             await self.drop()
 
     async def volume_receive(self, requests, existing, device):
@@ -223,7 +293,7 @@ class _PgTransportBuffer(TransportBuffer):
             pg = await asyncio.get_running_loop().run_in_executor(
                 None, cache.connect, self.pair_info
             )
-            await _recv_all(pg, tensors, src=0)
+            await _recv_all(pg, tensors, src=0, tag=self.op_tag)
         # move to the store device if needed
         final = []
         for v in out:
@@ -237,10 +307,7 @@ class _PgTransportBuffer(TransportBuffer):
     async def get(self, requests: Sequence[Request]) -> List[Any]:
         volume = self._volume_ref.volume
         cache: PgClientCache = self._client_ctx.cache(PgClientCache)
-        info, _store, _pg = cache.get_or_create(
-            self._volume_ref.volume_id, self.backend
-        )
-        self.pair_info = info
+        entry = self._begin_op()
         recvs: List[Tuple[int, torch.Tensor]] = []
         payload: List[Tuple[str, Any]] = []
         for i, r in enumerate(requests):
@@ -265,15 +332,21 @@ class _PgTransportBuffer(TransportBuffer):
             pg = await asyncio.get_running_loop().run_in_executor(
                 None, cache.ensure_pg, self._volume_ref.volume_id
             )
-            await _recv_all(pg, [t for _, t in recvs], src=1)
+            await _recv_all(pg, [t for _, t in recvs], src=1, tag=self.op_tag)
 
-        recv_task = asyncio.create_task(receiver()) if recvs else None
+        ok = False
         try:
-            reply = await volume.get.call_one(
-                self, [r.meta_only() for r in requests]
-            )
-            if recv_task is not None:
-                await recv_task
+            async with self._op_serializer(entry):
+                recv_task = asyncio.create_task(receiver()) if recvs else None
+                try:
+                    reply = await volume.get.call_one(
+                        self, [r.meta_only() for r in requests]
+                    )
+                    if recv_task is not None:
+                        await recv_task
+                finally:
+                    if recv_task is not None and not recv_task.done():
+                        recv_task.cancel()
             out: List[Any] = []
             ri = 0
             for i, (r, (kind, value)) in enumerate(zip(requests, reply)):
@@ -285,10 +358,10 @@ class _PgTransportBuffer(TransportBuffer):
                 if stage is not r.tensor_val:
                     r.tensor_val.copy_(stage)
                 out.append(r.tensor_val)
+            ok = True
             return out
         finally:
-            if recv_task is not None and not recv_task.done():
-                recv_task.cancel()
+            self._end_op(entry, ok)
             await self.drop()
 
     async def volume_send(self, requests, values):
@@ -308,7 +381,7 @@ class _PgTransportBuffer(TransportBuffer):
             pg = await asyncio.get_running_loop().run_in_executor(
                 None, cache.connect, self.pair_info
             )
-            await _send_all(pg, to_send, 0)
+            await _send_all(pg, to_send, 0, tag=self.op_tag)
         return reply
 
     def client_complete_get(self, requests, reply):  # unused: get() overridden

@@ -247,7 +247,7 @@ class ShmVolumeCache(TransportCache):
         for k in [k for k in self.put_segments if k[1] == key]:
             desc, seg = self.put_segments.pop(k)
             self.desc_by_storage.pop(seg.untyped_storage().data_ptr(), None)
-        for k in [k for k in self.get_segments if k[0] == key]:
+        for k in [k for k in self.get_segments if k[1] == key]:
             desc, seg = self.get_segments.pop(k)
             self.desc_by_storage.pop(seg.untyped_storage().data_ptr(), None)
 
@@ -361,6 +361,13 @@ class ShmTransportBuffer(TransportBuffer):
             s.synchronize()
         return out
 
+    # -- client get (stage): stamp the client identity ---------------------
+    async def client_stage_get(self, requests: Sequence[Request]) -> None:
+        # get-response segments are keyed per client: a get concurrent with
+        # another client's get of the same key must never share a response
+        # segment (torn values for the slower reader otherwise)
+        self.client_uid = self._client_ctx.uid
+
     # -- volume get -------------------------------------------------------
     async def volume_send(self, requests, values):
         cache: ShmVolumeCache = self._volume_ctx.cache(ShmVolumeCache)
@@ -382,6 +389,7 @@ class ShmTransportBuffer(TransportBuffer):
                     reply.append(("seg", known.with_layout(v.dtype, v.shape)))
                     continue
             region = (
+                self.client_uid,
                 r.key,
                 r.tensor_slice.offsets if r.tensor_slice else None,
                 tuple(v.shape),

@@ -187,6 +187,13 @@ def _request_slice(value) -> Tuple[torch.Tensor, Optional[TensorSlice]]:
     return req.tensor_val, req.tensor_slice
 
 
+def _prod(shape) -> int:
+    n = 1
+    for s in shape:
+        n *= s
+    return n
+
+
 def _full_slice(shape) -> TensorSlice:
     shape = tuple(shape)
     return TensorSlice(
@@ -279,7 +286,10 @@ class DirectWeightSyncDest:
         self.key = key
         self._plan: Optional[List[_TransferOp]] = None
         self._handles: Optional[List[WeightHandle]] = None
-        self._plan_dest_id: Optional[int] = None
+        # STRONG reference to the dict the cached plan targets: `is` against
+        # a held reference can never alias a new dict (id() could, after GC
+        # reuses the address — the plan would then sync into orphaned tensors)
+        self._plan_dest: Optional[Dict[str, Any]] = None
 
     async def _fetch_handles(self) -> List[WeightHandle]:
         if self._handles is not None:
@@ -315,6 +325,7 @@ class DirectWeightSyncDest:
             if not srcs:
                 raise KeyError(f"source has no parameter {name!r}")
             covered: set = set()
+            covered_regions: List[Tuple[Tuple[int, ...], Tuple[int, ...]]] = []
             for h in srcs:
                 src_region = h.slice or _full_slice(h.desc.shape)
                 inter = src_region.intersect(wanted)
@@ -324,6 +335,7 @@ class DirectWeightSyncDest:
                 if region_key in covered:
                     continue  # replicated shard — read once
                 covered.add(region_key)
+                covered_regions.append(region_key)
                 if dest_local.dtype != h.desc.dtype:
                     raise TypeError(
                         f"{name}: dest dtype {dest_local.dtype} != "
@@ -380,6 +392,31 @@ class DirectWeightSyncDest:
                         scatter=(src_view, dst_view),
                     )
                 )
+            # the source shards' intersections must TILE the wanted region;
+            # a gap would silently leave stale weights in the generator's
+            # parameters (verify, don't trust the layout)
+            from torchstore_amd.ops.slicing import (
+                regions_disjoint,
+                union_volume,
+            )
+
+            want_vol = 1
+            for s in wanted.local_shape:
+                want_vol *= s
+            if not covered_regions:
+                got_vol = 0
+            elif regions_disjoint(covered_regions):
+                got_vol = sum(
+                    _prod(shape) for _off, shape in covered_regions
+                )
+            else:
+                got_vol = union_volume(covered_regions)
+            if got_vol < want_vol:
+                raise RuntimeError(
+                    f"direct sync plan for {name!r} covers {got_vol} of "
+                    f"{want_vol} destination elements — source shards do "
+                    "not tile the wanted region"
+                )
         return plan
 
     def invalidate(self) -> None:
@@ -387,10 +424,11 @@ class DirectWeightSyncDest:
         e.g. after a trainer restart)."""
         self._plan = None
         self._handles = None
+        self._plan_dest = None
 
     async def pull(self, dest_state_dict: Dict[str, Any]) -> None:
         codec = get_codec()
-        if self._plan_dest_id is not None and self._plan_dest_id != id(dest_state_dict):
+        if self._plan_dest is not None and self._plan_dest is not dest_state_dict:
             # a different destination dict: the cached plan points at the old
             # tensors — rebuild against the new ones
             self._plan = None
@@ -398,7 +436,7 @@ class DirectWeightSyncDest:
             handles = await self._fetch_handles()
             dest_flat, _ = _flatten(dest_state_dict)
             self._plan = self._build_plan(handles, dest_flat)
-            self._plan_dest_id = id(dest_state_dict)
+            self._plan_dest = dest_state_dict
             logger.info(
                 "direct sync plan: %d ops (%d zero-copy, %d pitched, %d recv)",
                 len(self._plan),
