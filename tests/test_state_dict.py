@@ -96,8 +96,14 @@ async def test_partial_push_invisible():
 async def test_strict_mismatch():
     async def body():
         await ts.put_state_dict({"a": torch.randn(4)}, "ck")
-        with pytest.raises(KeyError, match="not in the stored"):
+        # user-only entry: strict rejects
+        with pytest.raises(KeyError, match="mapping mismatch"):
             await ts.get_state_dict("ck", {"zz": torch.zeros(4)})
+        # stored-only entry (user dict is a subset): strict rejects too —
+        # the reference asserts full mapping EQUALITY both ways
+        await ts.put_state_dict({"a": torch.randn(4), "b": torch.randn(4)}, "ck2")
+        with pytest.raises(KeyError, match="mapping mismatch"):
+            await ts.get_state_dict("ck2", {"a": torch.zeros(4)})
         # non-strict: extra stored entries are fine, missing ones error later
         out = await ts.get_state_dict("ck", {"a": torch.zeros(4)}, strict=False)
         assert out["a"].abs().sum() > 0

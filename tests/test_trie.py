@@ -52,3 +52,35 @@ def test_overwrite_and_pop():
     assert t.pop("k", "d") == "d"
     with pytest.raises(KeyError):
         t.pop("k")
+
+
+def test_dotted_prefix_matching():
+    """Reference parity: pygtrie StringTrie(separator=".") filters dotted
+    module paths by component (torchstore storage_utils/trie.py)."""
+    t = Trie()
+    t["sd/model.layers.0.weight"] = 1
+    t["sd/model.layers.1.weight"] = 2
+    t["sd/model2.weight"] = 3
+    t["sd/<MAPPING>"] = 4
+    # dotted prefix descends "." components
+    assert sorted(t.keys_with_prefix("sd/model.layers")) == [
+        "sd/model.layers.0.weight",
+        "sd/model.layers.1.weight",
+    ]
+    # "sd/model" must NOT match "sd/model2" (component boundary)
+    assert sorted(t.keys_with_prefix("sd/model")) == [
+        "sd/model.layers.0.weight",
+        "sd/model.layers.1.weight",
+    ]
+    assert len(t.keys_with_prefix("sd")) == 4
+
+
+def test_mixed_separators_stay_distinct():
+    t = Trie()
+    t["a.b"] = 1
+    t["a/b"] = 2
+    assert t["a.b"] == 1
+    assert t["a/b"] == 2
+    assert len(t) == 2
+    del t["a.b"]
+    assert "a.b" not in t and t["a/b"] == 2

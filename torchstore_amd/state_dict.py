@@ -160,14 +160,18 @@ async def get_state_dict(
     if user_state_dict is not None:
         user_flat, user_mapping = _flatten(user_state_dict)
         if strict:
-            stored_keys = {
-                str(k) for k in mapping
-            }
+            # reference asserts MAPPING EQUALITY, both directions
+            # (torchstore state_dict_utils.py:146-152): entries the user
+            # has but the store doesn't AND entries the store has but the
+            # user doesn't are both strict-mode errors
+            stored_keys = {str(k) for k in mapping}
             missing = set(user_flat.keys()) - stored_keys
-            if missing:
+            extra = stored_keys - set(user_flat.keys())
+            if missing or extra:
                 raise KeyError(
-                    f"user state_dict has entries not in the stored one: "
-                    f"{sorted(missing)[:5]}..."
+                    f"strict get_state_dict mapping mismatch: "
+                    f"user-only entries {sorted(missing)[:5]}, "
+                    f"stored-only entries {sorted(extra)[:5]}"
                 )
         fetches = {f"{key}/{k}": v for k, v in user_flat.items()}
         chunks = await asyncio.gather(

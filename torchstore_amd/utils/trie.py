@@ -1,11 +1,15 @@
 """Separator-aware prefix trie used by the controller's key index.
 
 Replaces the reference's pygtrie dependency (torchstore
-``storage_utils/trie.py``) with a small self-contained implementation.
-Keys are strings split on a separator (default ``"/"``, matching the
-hierarchical ``"{key}/{flat_key}"`` layout the state_dict layer writes);
-``keys(prefix=...)`` matches whole path components, so prefix ``"model"``
-matches ``"model/layer.0"`` but not ``"model2"``.
+``storage_utils/trie.py``, a ``StringTrie`` with separator ``"."``) with a
+small self-contained implementation.  Keys are tokenized on BOTH ``"/"``
+(the hierarchical ``"{key}/{flat_key}"`` layout the state_dict layer
+writes) and ``"."`` (the dotted module paths inside a flat key), with each
+separator kept attached to the component it introduces — so ``"a.b"`` and
+``"a/b"`` are distinct keys, and ``keys(prefix=...)`` matches whole
+components across both conventions: prefix ``"sd/model"`` matches
+``"sd/model.layers.0.weight"`` but not ``"sd/model2"`` (the reference's
+dotted-prefix behavior users of ``ts.keys`` expect).
 """
 
 from __future__ import annotations
@@ -13,6 +17,7 @@ from __future__ import annotations
 from typing import Any, Dict, Iterator, List, MutableMapping, Optional, Tuple
 
 _MISSING = object()
+_SEPARATORS = ("/", ".")
 
 
 class _Node:
@@ -25,16 +30,28 @@ class _Node:
 
 
 class Trie(MutableMapping):
-    def __init__(self, separator: str = "/"):
-        self._sep = separator
+    def __init__(self, separators: Tuple[str, ...] = _SEPARATORS):
+        self._seps = frozenset(separators)
         self._root = _Node()
         self._len = 0
 
     # -- helpers ---------------------------------------------------------
     def _parts(self, key: str) -> List[str]:
+        """Tokenize with each separator ATTACHED to the part it opens:
+        ``"a/b.c"`` → ``["a", "/b", ".c"]`` — concatenation reconstructs
+        the exact key, and ``"a.b"`` / ``"a/b"`` stay distinct."""
         if not isinstance(key, str):
             raise TypeError(f"trie keys are strings, got {type(key)}")
-        return key.split(self._sep)
+        parts: List[str] = []
+        cur: List[str] = []
+        for ch in key:
+            if ch in self._seps and cur:
+                parts.append("".join(cur))
+                cur = [ch]
+            else:
+                cur.append(ch)
+        parts.append("".join(cur))
+        return parts
 
     def _find(self, key: str) -> Optional[_Node]:
         node = self._root
@@ -96,7 +113,7 @@ class Trie(MutableMapping):
 
     def _iter(self, node: _Node, parts: List[str]) -> Iterator[str]:
         if node.has_value:
-            yield self._sep.join(parts)
+            yield "".join(parts)
         for part in node.children:
             parts.append(part)
             yield from self._iter(node.children[part], parts)
