@@ -34,11 +34,11 @@ def main():
 
     # export cost (fresh allocations vs same block)
     t = torch.randn(1 << 20, device="cuda")
-    timeit("ipc_export same tensor", lambda: e.ipc_export(t.data_ptr(), 0))
+    timeit("ipc_export same tensor", lambda: e.ipc_export(t.data_ptr(), 0, 0))
 
     # can a process open its OWN handle? (decides bench --mode direct shape)
     try:
-        h, off = e.ipc_export(t.data_ptr(), 0)
+        h, off, _size = e.ipc_export(t.data_ptr(), 0, 0)
         base = e.ipc_open(bytes(h), 0, 0)
         probe = torch.empty_like(t)
         e.copy_batch([(probe.data_ptr(), 0, base + off, 0, t.numel() * 4)])
@@ -52,7 +52,7 @@ def main():
 
     def export_many():
         for x in ts:
-            e.ipc_export(x.data_ptr(), 0)
+            e.ipc_export(x.data_ptr(), 0, 0)
 
     dt = timeit("ipc_export x64 tensors", export_many, n=5)
     print(f"  per export: {dt/64*1e3:.3f} ms")

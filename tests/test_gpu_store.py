@@ -387,3 +387,29 @@ async def test_gpu_put_does_not_sync_foreign_stream():
         other.synchronize()
 
     await _with_store(body)
+
+
+@requires_gpu
+async def test_export_cache_survives_empty_cache():
+    """Free a stored block to the OS (empty_cache), re-allocate at (likely)
+    the same base, re-put: the export-handle cache must not serve the stale
+    handle — the generation guard flushes it (csrc ipc_export)."""
+
+    async def body():
+        a = torch.full((32 << 20,), 1.0, device="cuda")  # 128 MB block
+        await ts.put("gen/a", a)
+        got = await ts.get("gen/a")
+        assert torch.equal(got, a)
+        base_a = a.data_ptr()
+        del a, got
+        torch.cuda.empty_cache()  # returns the block to the OS
+        b = torch.full((32 << 20,), 2.0, device="cuda")
+        reused = b.data_ptr() == base_a  # usually true; test is strongest then
+        await ts.put("gen/b", b)
+        out = await ts.get("gen/b")
+        torch.cuda.synchronize()
+        assert torch.equal(out, b), (
+            f"stale IPC handle served old bytes (base reused={reused})"
+        )
+
+    await _with_store(body, transport=TransportType.HIP_IPC)

@@ -101,27 +101,39 @@ static void ensure_desc_capacity(DevicePool& p, int device, size_t bytes) {
 // Export handles are cached per allocator block (hipIpcGetMemHandle does a
 // dmabuf export ioctl — tens of ms — while hipMemGetAddressRange is cheap).
 // Same design as the reference's (data_ptr, nbytes)-keyed RdmaMemory cache
-// (torchstore torchcomms/cache.py:150-187).  The cache is only invalid if a
-// block is returned to the OS (torch.cuda.empty_cache) and the same base is
-// re-allocated — call ipc_export_cache_clear() around such events.
+// (torchstore torchcomms/cache.py:150-187).  A cached handle goes stale
+// only when its block is returned to the OS (torch.cuda.empty_cache /
+// OOM-retry release) and the same base address is re-allocated; callers
+// therefore pass an allocator GENERATION (the caching allocator's
+// segment-freed counter): a changed generation flushes that device's
+// cache before lookup — no manual clear call needed for correctness.
+struct ExportCache {
+  long long gen = -(1ll << 62);
+  std::unordered_map<uintptr_t, std::string> map;
+};
 static std::mutex g_export_mutex;
-static std::unordered_map<uintptr_t, std::string> g_export_cache;
+static std::unordered_map<int, ExportCache> g_export_caches;
 
 // Returns (handle_bytes, offset_in_block, block_size).  Callers MUST route
 // blocks >= 2 GiB through the chunked-staging path: hipIpcOpenMemHandle of
 // a >=2^31-byte dmabuf hangs on this platform (measured; the export itself
 // succeeds, the peer's import never returns).
-static py::tuple ipc_export(uintptr_t ptr, int device) {
+static py::tuple ipc_export(uintptr_t ptr, int device, long long generation) {
   HIP_CHECK(hipSetDevice(device));
   void* base = nullptr;
   size_t size = 0;
   HIP_CHECK(hipMemGetAddressRange(&base, &size, reinterpret_cast<void*>(ptr)));
   uintptr_t base_u = reinterpret_cast<uintptr_t>(base);
   std::string handle_str;
-  {
+  if (generation >= 0) {  // negative generation = caller opts out of caching
     std::lock_guard<std::mutex> lock(g_export_mutex);
-    auto it = g_export_cache.find(base_u);
-    if (it != g_export_cache.end()) {
+    ExportCache& c = g_export_caches[device];
+    if (generation != c.gen) {
+      c.map.clear();
+      c.gen = generation;
+    }
+    auto it = c.map.find(base_u);
+    if (it != c.map.end()) {
       handle_str = it->second;
     }
   }
@@ -129,8 +141,10 @@ static py::tuple ipc_export(uintptr_t ptr, int device) {
     hipIpcMemHandle_t handle;
     HIP_CHECK(hipIpcGetMemHandle(&handle, base));
     handle_str.assign(reinterpret_cast<const char*>(&handle), sizeof(handle));
-    std::lock_guard<std::mutex> lock(g_export_mutex);
-    g_export_cache.emplace(base_u, handle_str);
+    if (generation >= 0) {
+      std::lock_guard<std::mutex> lock(g_export_mutex);
+      g_export_caches[device].map.emplace(base_u, handle_str);
+    }
   }
   return py::make_tuple(py::bytes(handle_str),
                         static_cast<uint64_t>(ptr - base_u),
@@ -139,7 +153,7 @@ static py::tuple ipc_export(uintptr_t ptr, int device) {
 
 static void ipc_export_cache_clear() {
   std::lock_guard<std::mutex> lock(g_export_mutex);
-  g_export_cache.clear();
+  g_export_caches.clear();
 }
 
 static uintptr_t ipc_open(py::bytes handle_bytes, int local_device,
@@ -622,7 +636,8 @@ static void sync_device(int device) {
 
 PYBIND11_MODULE(_hipstore, m) {
   m.doc() = "torchstore_amd native core (HIP/CDNA4, gfx950)";
-  m.def("ipc_export", &ipc_export, py::arg("ptr"), py::arg("device"));
+  m.def("ipc_export", &ipc_export, py::arg("ptr"), py::arg("device"),
+        py::arg("generation"));
   m.def("ipc_export_cache_clear", &ipc_export_cache_clear);
   m.def("ipc_open", &ipc_open, py::arg("handle"), py::arg("local_device"),
         py::arg("src_device"));

@@ -247,8 +247,27 @@ def pack_region(src_view: torch.Tensor) -> torch.Tensor:
     return out
 
 
-def ipc_export(ptr: int, device_index: int):
-    return ext().ipc_export(ptr, device_index)
+def alloc_generation(device_index: int) -> int:
+    """Caching-allocator generation for the export-handle cache.
+
+    ``segment.all.freed`` increments exactly when an allocator block is
+    returned to the OS (``empty_cache`` / OOM-retry release) — the only
+    event after which a cached ``hipIpcMemHandle_t`` for a re-allocated
+    base address would be stale.  Batch call sites compute this once per
+    batch (``torch.cuda.memory_stats`` is ~100 µs, too slow per tensor).
+    """
+    try:
+        return int(
+            torch.cuda.memory_stats(device_index).get("segment.all.freed", 0)
+        )
+    except Exception:  # noqa: BLE001 — stats unavailable: disable caching
+        return -1
+
+
+def ipc_export(ptr: int, device_index: int, generation: Optional[int] = None):
+    if generation is None:
+        generation = alloc_generation(device_index)
+    return ext().ipc_export(ptr, device_index, generation)
 
 
 def ipc_open(handle: bytes, local_device: int, src_device: int) -> int:

@@ -77,9 +77,11 @@ class BlockTooLargeError(RuntimeError):
     pass
 
 
-def export_tensor(t: torch.Tensor) -> IpcDescriptor:
+def export_tensor(
+    t: torch.Tensor, generation: Optional[int] = None
+) -> IpcDescriptor:
     """Export; raises :class:`BlockTooLargeError` for ≥2 GiB blocks."""
-    desc = try_export(t)
+    desc = try_export(t, generation)
     if desc is None:
         raise BlockTooLargeError(
             f"tensor lives in a >=2GiB allocator block "
@@ -89,10 +91,19 @@ def export_tensor(t: torch.Tensor) -> IpcDescriptor:
     return desc
 
 
-def try_export(t: torch.Tensor) -> Optional[IpcDescriptor]:
+def try_export(
+    t: torch.Tensor, generation: Optional[int] = None
+) -> Optional[IpcDescriptor]:
+    """``generation`` is the allocator generation guarding the handle cache
+    (``ops.gpu.alloc_generation``); batch call sites compute it once per
+    device per batch, cold paths let it default."""
     assert t.is_contiguous() and t.device.type == "cuda"
+    from torchstore_amd.ops import gpu
+
+    if generation is None:
+        generation = gpu.alloc_generation(t.device.index)
     handle, offset, block_size = _ext().ipc_export(
-        t.data_ptr(), t.device.index
+        t.data_ptr(), t.device.index, generation
     )
     if block_size >= IPC_BLOCK_LIMIT:
         return None
@@ -329,10 +340,13 @@ class HipIpcTransportBuffer(TransportBuffer):
         # be visible before the volume's one-sided pulls read the staging
         # memory from another process — so the sync happens once per device
         # AFTER the whole staging loop, not at the first tensor seen
+        from torchstore_amd.ops import gpu as gpu_ops
+
+        gens = {di: gpu_ops.alloc_generation(di) for di in devices}
         for di in devices:
             torch.cuda.current_stream(torch.device("cuda", di)).synchronize()
         for i, tc in staged:
-            desc = try_export(tc)
+            desc = try_export(tc, gens[tc.device.index])
             if desc is None:
                 token = await self._chunked_put_windows(tc)
                 payload[i] = ("chunked", token)
