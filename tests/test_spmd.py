@@ -100,3 +100,91 @@ def test_spmd_env_parsing():
     assert env.rank == 3 and env.num_hosts == 4
     with pytest.raises(RuntimeError, match="missing"):
         SPMDEnv.from_env({})
+
+
+def _multihost_worker(rank, world, local_world, master_port, result_path,
+                      strategy_name):
+    """Each process fakes being a different HOST (HOSTNAME env) with
+    local_world ranks per host — exercises the per-host volume spawn path
+    (reference capability: Monarch host-mesh spawning, spmd.py:317-326)."""
+    host_index = rank // local_world
+    os.environ.update(
+        {
+            "RANK": str(rank),
+            "LOCAL_RANK": str(rank % local_world),
+            "WORLD_SIZE": str(world),
+            "LOCAL_WORLD_SIZE": str(local_world),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(master_port),
+            "HOSTNAME": f"fakehost{host_index}",
+        }
+    )
+    import torchstore_amd as ts
+
+    strategy = {
+        "local_rank": ts.LocalRankStrategy,
+        "host": ts.HostStrategy,
+    }[strategy_name]()
+
+    async def main():
+        controller = await ts.initialize_spmd(strategy=strategy)
+        c = ts.client()
+        volumes = await c._ensure_volumes()
+        hostnames = sorted({v.hostname for v in volumes.values()})
+        volume_ids = sorted(volumes.keys())
+        # every rank writes; every rank reads the peer host's key
+        await ts.put(f"r{rank}", torch.full((8,), float(rank)))
+        peer = (rank + local_world) % world  # a rank on the OTHER host
+        for _ in range(200):
+            if await ts.exists(f"r{peer}"):
+                break
+            await asyncio.sleep(0.05)
+        out = await ts.get(f"r{peer}")
+        ok = bool(out.eq(float(peer)).all())
+        await ts.shutdown()
+        return {
+            "ok": ok,
+            "hostnames": hostnames,
+            "volume_ids": volume_ids,
+        }
+
+    try:
+        result = asyncio.run(main())
+    except Exception as exc:  # noqa: BLE001
+        import traceback
+
+        result = {"error": f"{exc}\n{traceback.format_exc()}"}
+    with open(result_path, "w") as f:
+        json.dump(result, f)
+
+
+@pytest.mark.parametrize("strategy_name", ["host", "local_rank"])
+def test_spmd_multihost_volume_placement(strategy_name):
+    """2 fake hosts x 1 rank: each host's local-rank-0 spawns its own
+    volumes; the controller sees volumes on BOTH hostnames."""
+    world, local_world = 2, 1
+    port = pick_free_port()
+    ctx = mp.get_context("spawn")
+    procs, paths = [], []
+    for rank in range(world):
+        path = tempfile.mktemp(prefix=f"spmd-mh-{rank}")
+        paths.append(path)
+        p = ctx.Process(
+            target=_multihost_worker,
+            args=(rank, world, local_world, port, path, strategy_name),
+        )
+        p.start()
+        procs.append(p)
+    for p in procs:
+        p.join(timeout=240)
+        assert not p.is_alive(), "multihost spmd worker hung"
+    for rank, path in enumerate(paths):
+        with open(path) as f:
+            result = json.load(f)
+        assert "error" not in result, f"rank {rank}: {result.get('error')}"
+        assert result["ok"], f"rank {rank} read wrong peer data"
+        assert result["hostnames"] == ["fakehost0", "fakehost1"], result
+        if strategy_name == "host":
+            assert result["volume_ids"] == ["fakehost0", "fakehost1"]
+        else:
+            assert result["volume_ids"] == ["0", "1"]

@@ -60,10 +60,12 @@ class SPMDEnv:
 
 
 class _SPMDSession:
-    def __init__(self, env: SPMDEnv, store, store_name: str):
+    def __init__(self, env: SPMDEnv, store, store_name: str, local_mesh=None):
         self.env = env
         self.store = store
         self.store_name = store_name
+        # non-rank-0 hosts: the volume mesh this host's local-rank-0 spawned
+        self.local_mesh = local_mesh
 
     async def shutdown(self) -> None:
         import asyncio
@@ -96,6 +98,11 @@ class _SPMDSession:
             status = self.store.get(
                 f"torchstore_amd/{self.store_name}/shutdown"
             ).decode()
+            if self.local_mesh is not None:
+                # this host's volumes, spawned by its local-rank-0 — stopped
+                # after rank 0's teardown (which resets them over RPC) so
+                # the reset never races the process exit
+                await self.local_mesh.stop()
             self.store.add(done_key, 1)
             if status != "ok":
                 raise RuntimeError(f"rank 0 store teardown failed: {status}")
@@ -132,24 +139,88 @@ async def initialize_spmd(
         timeout=timedelta(seconds=timeout_s),
     )
     key = f"torchstore_amd/{store_name}/controller"
+    total_volumes = strategy.num_volumes_for(env.world_size, env.num_hosts)
+    per_host = max(1, total_volumes // env.num_hosts)
+    host_index = env.rank // max(1, env.local_world_size)
+    count_key = f"torchstore_amd/{store_name}/volumes_registered"
+    local_mesh = None
     if env.rank == 0:
-        num_volumes = strategy.num_volumes_for(env.world_size, env.num_hosts)
+        # rank 0 spawns the controller + ITS OWN host's volumes only; the
+        # other hosts' local-rank-0 processes spawn theirs below (the
+        # reference places volumes on every host via Monarch host meshes,
+        # torchstore spmd.py:317-326 — here each host self-spawns and
+        # registers with the controller)
         controller = await api.initialize(
-            num_storage_volumes=num_volumes,
+            num_storage_volumes=per_host,
             strategy=strategy,
             store_name=store_name,
             storage_device=storage_device,
         )
         store.set(key, pickle.dumps(controller))
+        store.add(count_key, per_host)
     else:
         controller = pickle.loads(store.get(key))
         api.attach(controller, strategy, store_name)
-    _spmd_sessions[store_name] = _SPMDSession(env, store, store_name)
+        if env.local_rank == 0 and host_index > 0:
+            local_mesh, infos = await _spawn_host_volumes(
+                strategy, store_name, storage_device, env, host_index,
+                per_host,
+            )
+            await controller.register_volumes.call_one(infos)
+            store.add(count_key, len(infos))
+    # everyone waits until every host's volumes are registered — a client
+    # resolving volumes before that would cache an incomplete list
+    while int(store.add(count_key, 0)) < total_volumes:
+        import asyncio
+
+        await asyncio.sleep(0.02)
+    _spmd_sessions[store_name] = _SPMDSession(env, store, store_name, local_mesh)
     logger.info(
-        "spmd store %s up: rank %d/%d attached", store_name, env.rank,
-        env.world_size,
+        "spmd store %s up: rank %d/%d attached (%d volumes, %d hosts)",
+        store_name, env.rank, env.world_size, total_volumes, env.num_hosts,
     )
     return controller
+
+
+async def _spawn_host_volumes(
+    strategy: PlacementStrategy,
+    store_name: str,
+    storage_device: str,
+    env: SPMDEnv,
+    host_index: int,
+    per_host: int,
+):
+    """Spawn this host's volume processes (called on local-rank-0 of every
+    non-zero host) and build their registration infos."""
+    import asyncio
+
+    from torchstore_amd.controller import VolumeInfo
+    from torchstore_amd.runtime import spawn_actors
+
+    seed = strategy.volume_id_seed
+    if seed == "rank":
+        seed = f"rank_offset:{host_index * env.local_world_size}"
+    mesh = await asyncio.to_thread(
+        spawn_actors,
+        per_host,
+        _volume_cls(),
+        f"{store_name}-volume-h{host_index}",
+        volume_id_seed=seed,
+        device=storage_device,
+        timeout=240.0,
+    )
+    ids = await mesh.get_id.call()
+    infos = [
+        VolumeInfo(volume_id=vid, hostname=host, device=dev, handle=h)
+        for (vid, host, dev), h in zip(ids, mesh.handles)
+    ]
+    return mesh, infos
+
+
+def _volume_cls():
+    from torchstore_amd.storage import StorageVolume
+
+    return StorageVolume
 
 
 async def shutdown_spmd(store_name: str = api.DEFAULT_STORE) -> None:

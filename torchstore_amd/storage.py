@@ -231,6 +231,10 @@ class StorageVolume(Actor):
     def _make_volume_id(seed: str) -> str:
         if seed == "rank":
             return str(actor_context().rank)
+        if seed.startswith("rank_offset:"):
+            # multi-host spawning: host h's local spawn mesh restarts ranks
+            # at 0, the offset makes volume ids global (h * local_world)
+            return str(actor_context().rank + int(seed.split(":", 1)[1]))
         if seed == "host":
             return os.environ.get("HOSTNAME") or socket.gethostname()
         return seed
