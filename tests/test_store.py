@@ -235,3 +235,51 @@ async def test_transport_strategy_matrix(strategy_name, num_volumes, matrix_tran
         assert not await ts.exists("m/x")
     finally:
         await ts.shutdown()
+
+
+async def test_get_batch_plan_cache():
+    """Steady-state loops reuse the fetch plan (one verify RPC instead of
+    locate + planning); any layout change invalidates it."""
+
+    async def body():
+        from torchstore_amd.types import LocalShard, TensorSlice
+
+        a = torch.randn(64, 8)
+        b = torch.randn(32)
+        await ts.put_batch({"pc/a": a, "pc/b": b})
+        da, db = torch.zeros_like(a), torch.zeros_like(b)
+        c = ts.client()
+        fetches = {"pc/a": da, "pc/b": db}
+        out = await c.get_batch(fetches)
+        assert torch.equal(da, a) and torch.equal(db, b)
+        assert len(c._plan_cache) == 1
+        # same dests again: cached plan path
+        a2 = torch.randn(64, 8)
+        await ts.put("pc/a", a2)  # same layout -> fingerprint unchanged
+        out = await c.get_batch(fetches)
+        assert torch.equal(da, a2)
+        assert len(c._plan_cache) == 1
+        # different dest objects: plan must NOT be reused
+        da3 = torch.zeros_like(a)
+        await c.get_batch({"pc/a": da3, "pc/b": db})
+        assert torch.equal(da3, a2)
+        # layout change (plain tensor -> sharded) invalidates the fingerprint
+        shard0 = LocalShard(
+            tensor=a2[:32].clone(),
+            slice=TensorSlice((0, 0), (32, 8), (64, 8), (0,), (2,)),
+        )
+        shard1 = LocalShard(
+            tensor=a2[32:].clone(),
+            slice=TensorSlice((32, 0), (32, 8), (64, 8), (1,), (2,)),
+        )
+        await ts.put("pc/a", shard0)
+        await ts.put("pc/a", shard1)
+        da.zero_()
+        await c.get_batch(fetches)  # replans, still correct
+        assert torch.equal(da, a2)
+        # delete invalidates too: cached entry must not mask the KeyError
+        await ts.delete("pc/b")
+        with pytest.raises(KeyError):
+            await c.get_batch(fetches)
+
+    await _with_store(TransportType.RPC, body)

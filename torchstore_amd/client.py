@@ -46,6 +46,23 @@ class _SubFetch:
     result: Any = None
 
 
+@dataclass
+class _CachedFetchPlan:
+    """A reusable get_batch plan for steady-state sync loops.
+
+    Valid while (a) the caller passes the SAME destination objects (checked
+    by identity against held strong references) and (b) the controller
+    confirms every key's layout fingerprint is unchanged — one tiny RPC
+    instead of locate + slice planning (~2 ms/step for a Llama-8B dict).
+    """
+
+    likes: Dict[str, Any]
+    fingerprints: Dict[str, str]
+    plans: Dict[str, List[Tuple[str, "_SubFetch"]]]
+    by_volume: Dict[str, List["_SubFetch"]]
+    busy: bool = False  # a concurrent identical call must not share subs
+
+
 def _full_region_slice(global_shape: Sequence[int]) -> TensorSlice:
     shape = tuple(global_shape)
     return TensorSlice(
@@ -67,6 +84,7 @@ class LocalClient:
         self._strategy = strategy
         self._ctx = TransportContext()
         self._volumes: Optional[Dict[str, VolumeInfo]] = None
+        self._plan_cache: Dict[Tuple[str, ...], _CachedFetchPlan] = {}
 
     # -- bring-up ---------------------------------------------------------
     async def _ensure_volumes(self) -> Dict[str, VolumeInfo]:
@@ -121,6 +139,25 @@ class LocalClient:
             return {}
         tracker = LatencyTracker(f"get_batch[{len(fetches)}]")
         await self._ensure_volumes()
+        cache_key = tuple(sorted(fetches.keys()))
+        cached = self._plan_cache.get(cache_key)
+        if cached is not None and not cached.busy:
+            if all(cached.likes[k] is fetches[k] for k in fetches):
+                ok = await self._controller.verify_layouts.call_one(
+                    cached.fingerprints
+                )
+                if ok:
+                    tracker.step("verify")
+                    cached.busy = True
+                    try:
+                        return await self._run_plan(
+                            fetches, cached.plans, cached.by_volume, tracker
+                        )
+                    finally:
+                        cached.busy = False
+            del self._plan_cache[cache_key]
+        from torchstore_amd.controller import layout_fingerprint
+
         located = await self._controller.locate.call_one(list(fetches.keys()))
         tracker.step("locate")
         plans = {
@@ -133,6 +170,32 @@ class LocalClient:
         for subs in plans.values():
             for volume_id, sf in subs:
                 by_volume.setdefault(volume_id, []).append(sf)
+        # plans are reusable only when every destination came from the
+        # caller (is not None): plan-owned dests would be returned shared
+        # across calls otherwise
+        if all(like is not None for like in fetches.values()):
+            if len(self._plan_cache) >= 64:  # bound stale plan growth
+                self._plan_cache.pop(next(iter(self._plan_cache)))
+            self._plan_cache[cache_key] = _CachedFetchPlan(
+                likes=dict(fetches),
+                fingerprints={
+                    key: layout_fingerprint(located[key]) for key in fetches
+                },
+                plans=plans,
+                by_volume=by_volume,
+            )
+        return await self._run_plan(fetches, plans, by_volume, tracker)
+
+    async def _run_plan(
+        self,
+        fetches: Dict[str, Any],
+        plans: Dict[str, List[Tuple[str, _SubFetch]]],
+        by_volume: Dict[str, List[_SubFetch]],
+        tracker: LatencyTracker,
+    ) -> Dict[str, Any]:
+        for subs in by_volume.values():
+            for sf in subs:
+                sf.result = None
         await asyncio.gather(
             *(self._fetch_volume(vid, sfs) for vid, sfs in by_volume.items())
         )
@@ -359,4 +422,5 @@ class LocalClient:
         return await self._controller.key_exists.call_one(key)
 
     def close(self) -> None:
+        self._plan_cache.clear()
         self._ctx.close()

@@ -65,6 +65,24 @@ class VolumeInfo:
     handle: ActorHandle
 
 
+def layout_fingerprint(locations: Dict[str, StorageInfo]) -> str:
+    """Deterministic digest of a key's placement (volumes + slices).
+
+    Clients cache fetch plans keyed on this: one cheap ``verify_layouts``
+    RPC replaces locate + replanning in steady-state sync loops; any
+    layout change (re-shard, delete, volume move) mismatches and forces a
+    replan.  Built from sorted reprs — never the salted builtin hash.
+    """
+    import hashlib
+
+    parts = []
+    for vid in sorted(locations):
+        info = locations[vid]
+        slices = ";".join(sorted(repr(s) for s in info.tensor_slices))
+        parts.append(f"{vid}|{info.object_type.value}|{slices}")
+    return hashlib.md5("\n".join(parts).encode()).hexdigest()
+
+
 def _info_from_request(request: Request) -> StorageInfo:
     if request.is_object:
         return StorageInfo(ObjectType.OBJECT)
@@ -173,6 +191,17 @@ class Controller(Actor):
                 )
             out[key] = locations
         return out
+
+    @endpoint
+    def verify_layouts(self, fingerprints: Dict[str, str]) -> bool:
+        """True iff every key still has exactly the fingerprinted layout."""
+        for key, fp in fingerprints.items():
+            locations = self.index.get(key)
+            if locations is None or not self._is_fully_committed(locations):
+                return False
+            if layout_fingerprint(locations) != fp:
+                return False
+        return True
 
     @endpoint
     def notify_delete(self, key: str, missing_ok: bool = False) -> List[str]:
