@@ -361,6 +361,36 @@ async def test_direct_weight_sync_real_ipc():
         assert torch.equal(
             cast_dst["master"].cpu(), snap3["master"].to(torch.bfloat16)
         )
+
+        # partial-overlap dest (rows 128..384 of w): the plan's "2d"
+        # pitched ops run as STRIDED REMOTE KERNEL READS — mutate + re-pull
+        # proves the system-scope acquire keeps them coherent (no stale
+        # locally-cached lines from the previous pull)
+        from torchstore_amd.types import LocalShard, TensorSlice
+        from torchstore_amd.weight_sync import DirectWeightSyncDest
+
+        shard_dst = {
+            "w": LocalShard(
+                tensor=torch.zeros(256, 512, device="cuda"),
+                slice=TensorSlice(
+                    offsets=(128, 0), local_shape=(256, 512),
+                    global_shape=(512, 512), coordinates=(0,),
+                    mesh_shape=(1,),
+                ),
+            ),
+        }
+        dest2 = DirectWeightSyncDest(ts.client(), "dsync")
+        pull_sd = {"w": shard_dst["w"]}  # same dict → cached plan reused
+        await dest2.pull(pull_sd)
+        torch.cuda.synchronize()
+        snap4 = await src.snapshot.call_one()
+        assert torch.equal(shard_dst["w"].tensor.cpu(), snap4["w"][128:384])
+        await src.mutate.call_one()
+        await dest2.pull(pull_sd)
+        torch.cuda.synchronize()
+        snap5 = await src.snapshot.call_one()
+        assert not torch.equal(snap4["w"], snap5["w"])
+        assert torch.equal(shard_dst["w"].tensor.cpu(), snap5["w"][128:384])
     finally:
         if mesh is not None:
             await mesh.stop()

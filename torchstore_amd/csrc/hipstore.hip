@@ -274,9 +274,24 @@ __device__ __forceinline__ void row_offsets(const SliceDesc& d, uint64_t row,
 // runs 4 independent waves, so rows never serialize behind each other
 // inside a block, and the x4-unrolled 16B path keeps 4 loads in flight
 // per lane (64 lanes x 16B x 4 = 4 KiB outstanding per wave).
+//
+// NT: non-temporal stores in the 16B path — a bulk copy's destination is
+// never re-read by this kernel, so dropping the lines from L2 leaves the
+// cache to traffic that can reuse it (toggle measured on hardware).
+// REMOTE: sources are IPC-mapped PEER memory (one-sided batched reads over
+// xGMI, replacing per-piece SDMA enqueues); lines from a previous pull may
+// be cached locally, so each workgroup issues one system-scope acquire
+// before reading (validated by the cross-process staleness GPU test).
+template <bool NT, bool REMOTE>
 __global__ void __launch_bounds__(256)
 copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
                    uint64_t total_units, uint32_t tile_bytes) {
+  if (REMOTE) {
+    if (threadIdx.x == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+    }
+    __syncthreads();
+  }
   const uint32_t wave = threadIdx.x >> 6;
   const uint32_t lane = threadIdx.x & 63u;
   const uint64_t stride = (uint64_t)gridDim.x * 4;
@@ -316,14 +331,28 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
         uint4 v5 = s4[i + 320];
         uint4 v6 = s4[i + 384];
         uint4 v7 = s4[i + 448];
-        d4[i] = v0;
-        d4[i + 64] = v1;
-        d4[i + 128] = v2;
-        d4[i + 192] = v3;
-        d4[i + 256] = v4;
-        d4[i + 320] = v5;
-        d4[i + 384] = v6;
-        d4[i + 448] = v7;
+        if (NT) {
+          // uint4 is a class; nontemporal builtins need a real vector type
+          typedef uint32_t v4u __attribute__((ext_vector_type(4)));
+          v4u* dv = reinterpret_cast<v4u*>(d4);
+          __builtin_nontemporal_store(*reinterpret_cast<v4u*>(&v0), dv + i);
+          __builtin_nontemporal_store(*reinterpret_cast<v4u*>(&v1), dv + i + 64);
+          __builtin_nontemporal_store(*reinterpret_cast<v4u*>(&v2), dv + i + 128);
+          __builtin_nontemporal_store(*reinterpret_cast<v4u*>(&v3), dv + i + 192);
+          __builtin_nontemporal_store(*reinterpret_cast<v4u*>(&v4), dv + i + 256);
+          __builtin_nontemporal_store(*reinterpret_cast<v4u*>(&v5), dv + i + 320);
+          __builtin_nontemporal_store(*reinterpret_cast<v4u*>(&v6), dv + i + 384);
+          __builtin_nontemporal_store(*reinterpret_cast<v4u*>(&v7), dv + i + 448);
+        } else {
+          d4[i] = v0;
+          d4[i + 64] = v1;
+          d4[i + 128] = v2;
+          d4[i + 192] = v3;
+          d4[i + 256] = v4;
+          d4[i + 320] = v5;
+          d4[i + 384] = v6;
+          d4[i + 448] = v7;
+        }
       }
       for (; i + 192 < n4; i += 256) {
         uint4 a = s4[i];
@@ -352,10 +381,18 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
   }
 }
 
+static bool use_nt_stores() {
+  static int v = [] {
+    const char* e = getenv("HIPSTORE_NT");
+    return e ? atoi(e) : 1;  // default decided by the hardware microbench
+  }();
+  return v != 0;
+}
+
 // upload descriptors through the pinned staging buffer + launch the kernel
 static void launch_slice_descs(std::vector<SliceDesc>& descs, uint64_t units,
                                int device, hipStream_t stream,
-                               uint32_t tile) {
+                               uint32_t tile, bool remote = false) {
   if (units == 0 || descs.empty()) return;
   DevicePool& p = pool_for(device);
   HIP_CHECK(hipSetDevice(device));
@@ -375,7 +412,12 @@ static void launch_slice_descs(std::vector<SliceDesc>& descs, uint64_t units,
   // memory-bound: cap grid at 2048 blocks, grid-stride the rest (guide
   // G11); each block consumes 4 units (one per wave)
   uint32_t grid = (uint32_t)std::min<uint64_t>((units + 3) / 4, 2048);
-  hipLaunchKernelGGL(copy_slices_kernel, dim3(grid), dim3(256), 0, stream,
+  auto kern = use_nt_stores()
+                  ? (remote ? copy_slices_kernel<true, true>
+                            : copy_slices_kernel<true, false>)
+                  : (remote ? copy_slices_kernel<false, true>
+                            : copy_slices_kernel<false, false>);
+  hipLaunchKernelGGL(kern, dim3(grid), dim3(256), 0, stream,
                      reinterpret_cast<const SliceDesc*>(p.d_desc),
                      (uint32_t)descs.size(), units, tile);
   HIP_CHECK(hipGetLastError());
@@ -457,7 +499,8 @@ using PySlice = std::tuple<uintptr_t, uintptr_t, uint64_t,
                            std::vector<int64_t>>;
 
 static void copy_slices(const std::vector<PySlice>& slices, int device,
-                        uintptr_t stream_handle, bool blocking) {
+                        uintptr_t stream_handle, bool blocking,
+                        bool remote) {
   if (slices.empty()) return;
   size_t n = slices.size();
   uint64_t total_bytes = 0;
@@ -497,7 +540,8 @@ static void copy_slices(const std::vector<PySlice>& slices, int device,
   if (units == 0) return;
 
   launch_slice_descs(descs, units, device,
-                     reinterpret_cast<hipStream_t>(stream_handle), tile);
+                     reinterpret_cast<hipStream_t>(stream_handle), tile,
+                     remote);
   if (blocking) {
     HIP_CHECK(
         hipStreamSynchronize(reinterpret_cast<hipStream_t>(stream_handle)));
@@ -648,6 +692,7 @@ PYBIND11_MODULE(_hipstore, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("copy_slices", &copy_slices, py::arg("slices"), py::arg("device"),
         py::arg("stream"), py::arg("blocking") = true,
+        py::arg("remote") = false,
         py::call_guard<py::gil_scoped_release>());
   m.def("cast_copy", &cast_copy, py::arg("src"), py::arg("src_dtype"),
         py::arg("dst"), py::arg("dst_dtype"), py::arg("numel"),

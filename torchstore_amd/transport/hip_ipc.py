@@ -144,34 +144,51 @@ class IpcOpenCache(TransportCache):
 
 
 class ChunkStagingCache(TransportCache):
-    """Volume-side pool of <2 GiB staging chunks for the windowed path."""
+    """Volume-side pool of <2 GiB staging chunks for the windowed path.
+
+    An operation acquires TWO chunks so the client can double-buffer:
+    window N+1's xGMI copy overlaps window N's commit RPC (the round-1
+    serial path lost ~25% bandwidth at 2 GiB to the per-window RPC).
+    """
 
     def __init__(self):
         self.free: List[Tuple[torch.Tensor, IpcDescriptor]] = []
-        # token -> (staging tensor, staging desc, payload tensor)
-        self.by_token: Dict[str, Tuple[torch.Tensor, IpcDescriptor, torch.Tensor]] = {}
+        # token -> (list of (staging, desc), payload tensor)
+        self.by_token: Dict[
+            str, Tuple[List[Tuple[torch.Tensor, IpcDescriptor]], torch.Tensor]
+        ] = {}
 
-    def acquire(self, token: str, payload: torch.Tensor, device: torch.device):
-        if self.free:
-            staging, desc = self.free.pop()
-        else:
-            staging = torch.empty(CHUNK_BYTES, dtype=torch.uint8, device=device)
-            desc = export_tensor(staging)
-        self.by_token[token] = (staging, desc, payload)
-        return desc
+    def acquire(
+        self,
+        token: str,
+        payload: torch.Tensor,
+        device: torch.device,
+        nchunks: int = 2,
+    ) -> List[IpcDescriptor]:
+        chunks: List[Tuple[torch.Tensor, IpcDescriptor]] = []
+        for _ in range(nchunks):
+            if self.free:
+                chunks.append(self.free.pop())
+            else:
+                staging = torch.empty(
+                    CHUNK_BYTES, dtype=torch.uint8, device=device
+                )
+                chunks.append((staging, export_tensor(staging)))
+        self.by_token[token] = (chunks, payload)
+        return [desc for _s, desc in chunks]
 
     def payload(self, token: str) -> torch.Tensor:
-        return self.by_token[token][2]
+        return self.by_token[token][1]
 
-    def staging(self, token: str) -> torch.Tensor:
-        return self.by_token[token][0]
+    def staging(self, token: str, idx: int = 0) -> torch.Tensor:
+        return self.by_token[token][0][idx][0]
 
     def release(self, token: str) -> Optional[torch.Tensor]:
         entry = self.by_token.pop(token, None)
         if entry is None:
             return None
-        staging, desc, payload = entry
-        self.free.append((staging, desc))
+        chunks, payload = entry
+        self.free.extend(chunks)
         return payload
 
     def drop_key(self, key: str) -> None:
@@ -215,30 +232,46 @@ class HipIpcTransportBuffer(TransportBuffer):
 
     # -- chunked windows (client side) -----------------------------------
     async def _chunked_put_windows(self, t: torch.Tensor) -> str:
-        """Stream a big tensor into volume staging, window by window."""
+        """Stream a big tensor into volume staging, window by window.
+
+        Double-buffered over two staging chunks: while window N's commit
+        RPC is in flight, window N+1's xGMI copy runs concurrently (in an
+        executor thread — the C++ copy drops the GIL)."""
         volume = self._volume_ref.volume
         cache: IpcOpenCache = self._client_ctx.cache(IpcOpenCache)
         token = uuid.uuid4().hex
-        staging_desc = await volume.handshake.call_one(
+        staging_descs = await volume.handshake.call_one(
             self, (token, tuple(t.shape), t.dtype), "chunk_put_init"
         )
         try:
-            staging_ptr = cache.resolve(staging_desc, t.device.index)
+            ptrs = [
+                cache.resolve(d, t.device.index) for d in staging_descs
+            ]
             nbytes = t.numel() * t.element_size()
             base = t.data_ptr()
-            off = 0
-            while off < nbytes:
-                win = min(CHUNK_BYTES, nbytes - off)
-                _run_copies(
-                    [(staging_ptr, staging_desc.device_index,
-                      base + off, t.device.index, win)]
+            windows = [
+                (off, min(CHUNK_BYTES, nbytes - off))
+                for off in range(0, nbytes, CHUNK_BYTES)
+            ]
+            commit_task = None
+            for i, (off, win) in enumerate(windows):
+                c = i % len(ptrs)
+                await asyncio.to_thread(
+                    _run_copies,
+                    [(ptrs[c], staging_descs[c].device_index,
+                      base + off, t.device.index, win)],
                 )
-                await volume.handshake.call_one(
-                    self, (token, off, win), "chunk_put_commit"
+                if commit_task is not None:
+                    await commit_task  # chunk c is free again after this
+                commit_task = asyncio.ensure_future(
+                    volume.handshake.call_one(
+                        self, (token, c, off, win), "chunk_put_commit"
+                    )
                 )
-                off += win
+            if commit_task is not None:
+                await commit_task
         except BaseException:
-            # abort: return the staging chunk to the pool (uniflow's abort
+            # abort: return the staging chunks to the pool (uniflow's abort
             # phase — a failed transfer must not leak volume resources)
             try:
                 await volume.handshake.call_one(self, token, "chunk_release")
@@ -250,27 +283,54 @@ class HipIpcTransportBuffer(TransportBuffer):
     async def _chunked_get_windows(
         self, request: Request, dest: torch.Tensor
     ) -> str:
+        """Windowed fetch, double-buffered: window N's xGMI copy out of one
+        staging chunk overlaps window N+1's fill RPC into the other."""
         volume = self._volume_ref.volume
         cache: IpcOpenCache = self._client_ctx.cache(IpcOpenCache)
         token = uuid.uuid4().hex
-        staging_desc = await volume.handshake.call_one(
+        staging_descs = await volume.handshake.call_one(
             self, (token, request.meta_only()), "chunk_get_init"
         )
         try:
-            staging_ptr = cache.resolve(staging_desc, dest.device.index)
+            ptrs = [
+                cache.resolve(d, dest.device.index) for d in staging_descs
+            ]
             nbytes = dest.numel() * dest.element_size()
             base = dest.data_ptr()
-            off = 0
-            while off < nbytes:
-                win = min(CHUNK_BYTES, nbytes - off)
-                await volume.handshake.call_one(
-                    self, (token, off, win), "chunk_get_fill"
+            windows = [
+                (off, min(CHUNK_BYTES, nbytes - off))
+                for off in range(0, nbytes, CHUNK_BYTES)
+            ]
+            fill_task = None
+            prev: Optional[Tuple[int, int, int]] = None  # (chunk, off, win)
+            for i, (off, win) in enumerate(windows):
+                c = i % len(ptrs)
+                if fill_task is not None:
+                    await fill_task
+                fill_task = asyncio.ensure_future(
+                    volume.handshake.call_one(
+                        self, (token, c, off, win), "chunk_get_fill"
+                    )
                 )
-                _run_copies(
-                    [(base + off, dest.device.index,
-                      staging_ptr, staging_desc.device_index, win)]
+                if prev is not None:
+                    pc, poff, pwin = prev
+                    # copy the PREVIOUS (already filled) window while the
+                    # current fill RPC runs — different chunks, no overlap
+                    await asyncio.to_thread(
+                        _run_copies,
+                        [(base + poff, dest.device.index,
+                          ptrs[pc], staging_descs[pc].device_index, pwin)],
+                    )
+                prev = (c, off, win)
+            if fill_task is not None:
+                await fill_task
+            if prev is not None:
+                pc, poff, pwin = prev
+                await asyncio.to_thread(
+                    _run_copies,
+                    [(base + poff, dest.device.index,
+                      ptrs[pc], staging_descs[pc].device_index, pwin)],
                 )
-                off += win
         except BaseException:
             try:
                 await volume.handshake.call_one(self, token, "chunk_release")
@@ -288,8 +348,8 @@ class HipIpcTransportBuffer(TransportBuffer):
             payload = torch.empty(shape, dtype=dtype, device=device)
             return cache.acquire(token, payload, device)
         if phase == "chunk_put_commit":
-            token, dst_off, length = args
-            staging = cache.staging(token)
+            token, chunk_idx, dst_off, length = args
+            staging = cache.staging(token, chunk_idx)
             payload = cache.payload(token)
             _run_copies(
                 [(payload.data_ptr() + dst_off, device.index,
@@ -305,8 +365,8 @@ class HipIpcTransportBuffer(TransportBuffer):
             torch.cuda.current_stream(device).synchronize()
             return cache.acquire(token, packed, device)
         if phase == "chunk_get_fill":
-            token, src_off, length = args
-            staging = cache.staging(token)
+            token, chunk_idx, src_off, length = args
+            staging = cache.staging(token, chunk_idx)
             payload = cache.payload(token)
             _run_copies(
                 [(staging.data_ptr(), device.index,

@@ -268,9 +268,19 @@ class _PgTransportBuffer(TransportBuffer):
                     await volume.put.call_one(
                         self, [r.meta_only() for r in requests]
                     )
-                finally:
+                except BaseException:
+                    # the RPC failed: the send task may be stuck waiting for
+                    # a rendezvous that will never complete — don't let its
+                    # (timeout) error mask the primary failure
                     if send_task is not None:
-                        await send_task
+                        send_task.cancel()
+                        try:
+                            await send_task
+                        except BaseException:  # noqa: BLE001
+                            pass
+                    raise
+                if send_task is not None:
+                    await send_task
             ok = True
         finally:
             self._end_op(entry, ok)

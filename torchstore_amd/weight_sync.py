@@ -92,7 +92,58 @@ class IpcMemoryCodec:
         return export_tensor(t)
 
     def read_batch(self, ops: Sequence[_TransferOp], ctx) -> None:
-        """Execute every read as one striped batch over the stream pool."""
+        """Execute every one-sided read of a pull as ONE kernel launch.
+
+        The kernel reads IPC-mapped source memory directly (peer loads
+        over xGMI for cross-GPU sources) and writes the — possibly
+        strided — destinations: at N=8 this replaces ~2.3k pitched SDMA
+        enqueues per rank (~20 µs host cost each) with a single dispatch,
+        and units from different peers are in flight concurrently across
+        the grid, so several xGMI links are driven by one launch.  A
+        system-scope acquire at workgroup start keeps re-pulls coherent
+        (remote lines cached by a previous pull).  Set
+        ``TORCHSTORE_AMD_DIRECT_SDMA=1`` for the per-op SDMA fallback.
+        """
+        from torchstore_amd.transport.hip_ipc import IpcOpenCache
+        from torchstore_amd.ops import gpu
+
+        if os.environ.get("TORCHSTORE_AMD_DIRECT_SDMA", "0") == "1":
+            return self._read_batch_sdma(ops, ctx)
+        cache: IpcOpenCache = ctx.cache(IpcOpenCache)
+        descs = []
+        device = None
+        for op in ops:
+            device = op.dst.device
+            src_ptr = cache.resolve(op.desc, device.index)
+            if op.kind == "2d":
+                es = op.dst.element_size()
+                dpitch = (
+                    op.dst.stride(0) * es if op.dst.dim() == 2 else op.width
+                )
+                src = src_ptr + op.src_offset_bytes
+                if op.height == 1:
+                    descs.append((src, op.dst.data_ptr(), op.width, [], [], []))
+                else:
+                    descs.append(
+                        (src, op.dst.data_ptr(), op.width,
+                         [op.height], [op.spitch], [dpitch])
+                    )
+            else:
+                # flat read; row_bytes is u32 in the kernel — span in <=1 GiB
+                nbytes = op.desc.nbytes
+                dstp = op.dst.data_ptr()
+                off = 0
+                while nbytes - off > 0:
+                    n = min(nbytes - off, 1 << 30)
+                    descs.append((src_ptr + off, dstp + off, n, [], [], []))
+                    off += n
+        if descs:
+            gpu.ext().copy_slices(
+                descs, device.index, gpu._stream(device), True, True
+            )
+
+    def _read_batch_sdma(self, ops: Sequence[_TransferOp], ctx) -> None:
+        """Per-op SDMA path (pitched engines), striped over the stream pool."""
         from torchstore_amd.transport.hip_ipc import IpcOpenCache
         from torchstore_amd.ops import gpu
 
