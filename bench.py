@@ -174,6 +174,7 @@ async def run_bench(args, rank, world, local_rank, coord: Coord):
 
     gen_mesh = None
     pull_bytes = [0]
+    phases = {"put": 0.0, "barrier": 0.0, "get": 0.0}
     if args.mode == "direct":
         # trainer = this process; generator = a separate process on the same
         # GPU (IPC handles cannot be opened by their exporting process)
@@ -192,8 +193,13 @@ async def run_bench(args, rank, world, local_rank, coord: Coord):
         async def one_step():
             # push = staging refresh (no cast here → sync only);
             # pull = one batched one-sided read into generator memory
+            t0 = time.perf_counter()
             await ts.put_state_dict(src_sd, "bench", direct=True)
+            t1 = time.perf_counter()
             _dt, pull_bytes[0] = await gen.pull.call_one()
+            t2 = time.perf_counter()
+            phases["put"] += t1 - t0
+            phases["get"] += t2 - t1
 
     else:
         dst_sd = llama.make_local_shard_state_dict(
@@ -201,9 +207,16 @@ async def run_bench(args, rank, world, local_rank, coord: Coord):
         )
 
         async def one_step():
+            t0 = time.perf_counter()
             await ts.put_state_dict(src_sd, "bench")
+            t1 = time.perf_counter()
             coord.barrier()  # all shards committed before the reshard pull
+            t2 = time.perf_counter()
             await ts.get_state_dict("bench", dst_sd)
+            t3 = time.perf_counter()
+            phases["put"] += t1 - t0
+            phases["barrier"] += t2 - t1
+            phases["get"] += t3 - t2
 
     for _ in range(args.warmup):
         await one_step()
@@ -237,11 +250,24 @@ async def run_bench(args, rank, world, local_rank, coord: Coord):
                 f"({len(bad)} of {len(dst_sd)} entries wrong)"
             )
 
+    for k in phases:
+        phases[k] = 0.0  # drop warmup contributions
     t0 = time.perf_counter()
     for _ in range(args.steps):
         await one_step()
     barrier()
     elapsed = time.perf_counter() - t0
+
+    # per-rank phase breakdown (stderr): makes a bad SCALE curve diagnosable
+    import sys as _sys
+
+    print(
+        f"[bench rank {rank}/{world}] per-step phases: "
+        + " ".join(
+            f"{k}={v / args.steps * 1e3:.1f}ms" for k, v in phases.items()
+        ),
+        file=_sys.stderr, flush=True,
+    )
 
     times = coord.gather_floats(elapsed)
     if args.mode == "direct":

@@ -443,3 +443,94 @@ async def test_export_cache_survives_empty_cache():
         )
 
     await _with_store(body, transport=TransportType.HIP_IPC)
+
+
+async def test_packed_put_coalescing(monkeypatch):
+    """Put-side coalescing (the twin of the get bounce): small cross-device
+    tensors pack into one staging buffer — forced on one GPU by faking the
+    volume's device index. Mixed batch: packed, large (direct), strided
+    (kernel-inexpressible fallback), 0-d, objects, and overwrites."""
+    from torchstore_amd.transport.hip_ipc import HipIpcTransportBuffer
+
+    monkeypatch.setattr(
+        HipIpcTransportBuffer, "_volume_device_index", lambda self: 7
+    )
+
+    async def body():
+        big = torch.randn(20 << 20, device="cuda")  # 80 MB > threshold
+        strided = torch.randn(128, 64, device="cuda").t()  # stride(-1) != 1
+        items = {
+            "small0": torch.randn(300, 301, device="cuda"),
+            "small1": torch.randn(7, device="cuda", dtype=torch.bfloat16),
+            "zero_d": torch.tensor(3.5, device="cuda"),
+            "big": big,
+            "strided": strided,
+            "obj": {"meta": 42},
+        }
+        await ts.put_batch(items)
+        for k, v in items.items():
+            out = await ts.get(k)
+            if isinstance(v, torch.Tensor):
+                assert torch.equal(out, v.contiguous()), k
+            else:
+                assert out == v, k
+        # overwrite with same shapes: volume reuses stored tensors (prior)
+        items2 = {
+            "small0": torch.randn(300, 301, device="cuda"),
+            "small1": torch.randn(7, device="cuda", dtype=torch.bfloat16),
+        }
+        await ts.put_batch(items2)
+        for k, v in items2.items():
+            assert torch.equal(await ts.get(k), v), k
+
+    await _with_store(body, transport=TransportType.HIP_IPC)
+
+
+@requires_gpu
+async def test_fake8_reshard_topology(monkeypatch):
+    """The 8-rank fsdp→tp reshard flow dry-run on ONE GPU: 8 volumes (all
+    cuda:0), each 'rank' puts its Shard(0) slice, then pulls its TP slice
+    through the full multi-volume fan-out + commit gate + plan cache —
+    the code path the driver's 8-GPU SCALE run takes."""
+    import os
+
+    from torchstore_amd.models import llama
+    from torchstore_amd.strategy import LocalRankStrategy
+    from torchstore_amd.types import LocalShard
+
+    world, layers = 8, 2
+    await ts.initialize(
+        num_storage_volumes=world,
+        strategy=LocalRankStrategy(transport=TransportType.HIP_IPC),
+        storage_device="auto",
+    )
+    try:
+        for r in range(world):
+            monkeypatch.setenv("RANK", str(r))
+            src = llama.make_local_shard_state_dict(
+                r, world, llama.fsdp_placement, device="cuda:0",
+                layers=layers, pattern=True,
+            )
+            await ts.put_state_dict(src, "f8")
+        dsts = {}
+        for r in range(world):
+            monkeypatch.setenv("RANK", str(r))
+            dst = llama.make_local_shard_state_dict(
+                r, world, llama.tp_placement, device="cuda:0", layers=layers,
+            )
+            await ts.get_state_dict("f8", dst)
+            dsts[r] = dst
+        torch.cuda.synchronize()
+        for r, dst in dsts.items():
+            for name, v in dst.items():
+                local = v.tensor if isinstance(v, LocalShard) else v
+                offsets = (
+                    v.slice.offsets if isinstance(v, LocalShard)
+                    else (0,) * local.dim()
+                )
+                exp = llama.expected_pattern(
+                    tuple(local.shape), offsets, local.dtype, local.device
+                )
+                assert torch.equal(local, exp), f"rank {r} {name}"
+    finally:
+        await ts.shutdown()

@@ -213,9 +213,13 @@ class HipIpcTransportBuffer(TransportBuffer):
         super().__init__()
         # aligned with requests:
         #   ("ipc", IpcDescriptor) | ("inline", value) | ("chunked", token)
+        #   ("packed", (pack_idx, offset, shape, dtype)) — put coalescing
         #   get-side markers: ("fetch_obj"|"fetch_inline", None)
         self.payload: Optional[List[Tuple[str, Any]]] = None
         self.bounce_descs: List[IpcDescriptor] = []
+        # put-side coalescing: (desc, used_bytes) per client pack buffer
+        self.pack_descs: List[Tuple[IpcDescriptor, int]] = []
+        self._packs: List[torch.Tensor] = []
         self._hold: List[torch.Tensor] = []       # keep exports alive
         self._scratch: Dict[int, torch.Tensor] = {}  # req idx -> dense scratch
         self._bounces: List[torch.Tensor] = []
@@ -228,6 +232,7 @@ class HipIpcTransportBuffer(TransportBuffer):
         state["_hold"] = []
         state["_scratch"] = {}
         state["_bounces"] = []
+        state["_packs"] = []
         return state
 
     # -- chunked windows (client side) -----------------------------------
@@ -379,10 +384,17 @@ class HipIpcTransportBuffer(TransportBuffer):
         raise ValueError(f"unknown handshake phase {phase!r}")
 
     # ------------------------------------------------------------- put --
+    PACK_THRESHOLD = 32 << 20   # cross-device tensors below this coalesce
+    PACK_BUF_CAP = 512 << 20    # per pack buffer (stays exportable)
+
     async def client_stage_put(self, requests: Sequence[Request]) -> None:
+        from torchstore_amd.ops import gpu as gpu_ops
+
         payload: List[Tuple[str, Any]] = []
         staged: List[Tuple[int, torch.Tensor]] = []
+        pack_items: List[Tuple[int, torch.Tensor]] = []
         devices: set = set()
+        vol_dev = self._volume_device_index()
         for r in requests:
             if r.is_object:
                 payload.append(("inline", r.objects))
@@ -391,22 +403,103 @@ class HipIpcTransportBuffer(TransportBuffer):
             if t.device.type != "cuda":
                 payload.append(("inline", t))
                 continue
+            nbytes = t.numel() * t.element_size()
+            cross = vol_dev >= 0 and vol_dev != t.device.index
+            if cross and 0 < nbytes < self.PACK_THRESHOLD:
+                # put-side coalescing (the twin of the get-side bounce):
+                # small cross-device tensors pack into ONE staging buffer —
+                # one K1 pack launch here, one xGMI copy + one K2 scatter
+                # on the volume, instead of an SDMA enqueue per tensor
+                payload.append(("pending", None))
+                pack_items.append((len(payload) - 1, t))
+                devices.add(t.device.index)
+                continue
             tc = t.contiguous()
             self._hold.append(tc)
             devices.add(tc.device.index)
             payload.append(("pending", None))
             staged.append((len(payload) - 1, tc))
-        # every producing kernel AND every pack copy from .contiguous() must
-        # be visible before the volume's one-sided pulls read the staging
-        # memory from another process — so the sync happens once per device
-        # AFTER the whole staging loop, not at the first tensor seen
-        from torchstore_amd.ops import gpu as gpu_ops
 
+        self.pack_descs = []
+        self._packs: List[torch.Tensor] = []
+        pack_meta: List[Tuple[int, int, int]] = []  # (payload idx, pack idx, off)
+        if pack_items:
+            by_dev: Dict[int, List[Tuple[int, torch.Tensor]]] = {}
+            for i, t in pack_items:
+                by_dev.setdefault(t.device.index, []).append((i, t))
+            for dev, items in by_dev.items():
+                device = torch.device("cuda", dev)
+                buf = None
+                off = 0
+                copies = []  # (src_view, dst_ptr) for one batched launch
+                for i, t in items:
+                    nb = t.numel() * t.element_size()
+                    aligned = (nb + 255) & ~255
+                    if buf is None or off + aligned > buf.numel():
+                        if buf is not None:
+                            self.pack_descs[-1] = (self.pack_descs[-1][0], off)
+                        remaining = sum(
+                            (tt.numel() * tt.element_size() + 255) & ~255
+                            for j, tt in items
+                            if (j, tt) == (i, t) or j > i
+                        )
+                        size = min(self.PACK_BUF_CAP, max(remaining, aligned))
+                        buf = torch.empty(size, dtype=torch.uint8, device=device)
+                        self._packs.append(buf)
+                        self._hold.append(buf)
+                        self.pack_descs.append((None, 0))  # desc filled below
+                        off = 0
+                    copies.append((t, buf.data_ptr() + off))
+                    pack_meta.append((i, len(self._packs) - 1, off))
+                    payload[i] = (
+                        "packed",
+                        (len(self._packs) - 1, off, tuple(t.shape), t.dtype),
+                    )
+                    off += aligned
+                if buf is not None:
+                    self.pack_descs[-1] = (self.pack_descs[-1][0], off)
+                rejects = gpu_ops.copy_views_to_ptrs(
+                    copies, device, blocking=False
+                )
+                for t, _ptr in rejects:
+                    # kernel-inexpressible layout: pack via a contiguous copy
+                    tc = t.contiguous()
+                    idx = next(
+                        i for i, tt in pack_items if tt is t
+                    )
+                    pi, pk, poff = next(
+                        m for m in pack_meta if m[0] == idx
+                    )
+                    from torchstore_amd.ops.slicing import byte_view
+
+                    nb = tc.numel() * tc.element_size()
+                    self._packs[pk][poff : poff + nb].copy_(byte_view(tc))
+
+        # every producing kernel AND every pack copy must be visible before
+        # the volume's one-sided pulls read the staging memory from another
+        # process — so the sync happens once per device AFTER the whole
+        # staging loop, not at the first tensor seen
         gens = {di: gpu_ops.alloc_generation(di) for di in devices}
         for di in devices:
             torch.cuda.current_stream(torch.device("cuda", di)).synchronize()
+
+        resolved_packs: List[Tuple[Optional[IpcDescriptor], int]] = []
+        for buf, (_d, used) in zip(self._packs, self.pack_descs):
+            desc = try_export(buf, gens[buf.device.index])
+            resolved_packs.append((desc, used))
+        self.pack_descs = resolved_packs
+        for pi, pk, _off in pack_meta:
+            if self.pack_descs[pk][0] is None:
+                # pack buffer landed in an unexportable (>=2 GiB) block:
+                # degrade this entry to individual staging
+                t = requests[pi].tensor_val
+                tc = t.contiguous()
+                self._hold.append(tc)
+                torch.cuda.current_stream(tc.device).synchronize()
+                staged.append((pi, tc))
+
         for i, tc in staged:
-            desc = try_export(tc, gens[tc.device.index])
+            desc = try_export(tc, gens.get(tc.device.index))
             if desc is None:
                 token = await self._chunked_put_windows(tc)
                 payload[i] = ("chunked", token)
@@ -417,20 +510,28 @@ class HipIpcTransportBuffer(TransportBuffer):
     async def volume_receive(self, requests, existing, device):
         cache: IpcOpenCache = self._volume_ctx.cache(IpcOpenCache)
         chunks: ChunkStagingCache = self._volume_ctx.cache(ChunkStagingCache)
-        out: List[Any] = []
+        out: List[Any] = [None] * len(self.payload)
         copies: List[Tuple[int, int, int, int, int]] = []
-        for (kind, value), prior in zip(self.payload, existing):
+        # coalesced puts: (out idx, pack idx, off, shape, dtype, prior)
+        pack_entries: List[Tuple[int, int, int, Tuple, Any, Any]] = []
+        for i, ((kind, value), prior) in enumerate(
+            zip(self.payload, existing)
+        ):
             if kind == "inline":
                 if isinstance(value, torch.Tensor):
-                    out.append(value.to(device))
+                    out[i] = value.to(device)
                 else:
-                    out.append(value)
+                    out[i] = value
                 continue
             if kind == "chunked":
                 payload = chunks.release(value)
                 if payload is None:
                     raise RuntimeError("chunked put token unknown")
-                out.append(payload)
+                out[i] = payload
+                continue
+            if kind == "packed":
+                pk, off, shape, dtype = value
+                pack_entries.append((i, pk, off, shape, dtype, prior))
                 continue
             desc: IpcDescriptor = value
             src_ptr = cache.resolve(desc, device.index)
@@ -448,11 +549,48 @@ class HipIpcTransportBuffer(TransportBuffer):
                 (dst.data_ptr(), device.index, src_ptr, desc.device_index,
                  desc.nbytes)
             )
-            out.append(dst)
+            out[i] = dst
+        pack_locals: Dict[int, torch.Tensor] = {}
+        if pack_entries:
+            # ONE xGMI copy per pack buffer's used span, then one batched
+            # K2 scatter splitting it into stored tensors
+            for pk in {e[1] for e in pack_entries}:
+                desc, used = self.pack_descs[pk]
+                local = torch.empty(used, dtype=torch.uint8, device=device)
+                src_ptr = cache.resolve(desc, device.index)
+                copies.append(
+                    (local.data_ptr(), device.index, src_ptr,
+                     desc.device_index, used)
+                )
+                pack_locals[pk] = local
         if copies:
             # executor thread: the volume keeps serving other clients while
             # the batched pull runs (the C++ side drops the GIL)
             await asyncio.to_thread(_run_copies, copies)
+        if pack_entries:
+            from torchstore_amd.ops import gpu as gpu_ops
+
+            pairs = []
+            for i, pk, off, shape, dtype, prior in pack_entries:
+                local = pack_locals[pk]
+                numel = 1
+                for s in shape:
+                    numel *= s
+                nb = numel * torch._utils._element_size(dtype)
+                slot = local[off : off + nb].view(dtype).reshape(shape)
+                if (
+                    prior is not None
+                    and tuple(prior.shape) == tuple(shape)
+                    and prior.dtype == dtype
+                    and prior.is_contiguous()
+                    and prior.device == device
+                ):
+                    dst = prior
+                else:
+                    dst = torch.empty(shape, dtype=dtype, device=device)
+                pairs.append((slot, dst))
+                out[i] = dst
+            gpu_ops.copy_pairs(pairs, device, blocking=True)
         return out
 
     # ------------------------------------------------------------- get --
@@ -732,5 +870,7 @@ class HipIpcTransportBuffer(TransportBuffer):
         self._hold.clear()
         self._scratch.clear()
         self._bounces = []
+        self._packs = []
         self.bounce_descs = []
+        self.pack_descs = []
         self.payload = None
