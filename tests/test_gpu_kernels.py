@@ -114,3 +114,26 @@ def test_cast_tensor_api():
     b = cast_tensor(f, torch.bfloat16)
     torch.cuda.synchronize()
     assert torch.equal(b, f.to(torch.bfloat16))
+
+
+@requires_gpu
+def test_copy_pairs_row_packed_boundaries():
+    """Row-packed units (small aligned rows): row counts chosen to hit
+    partial last units and slice transitions inside one wave's chunk."""
+    import torch
+
+    from torchstore_amd.ops import gpu
+
+    torch.cuda.set_device(0)
+    pairs = []
+    expect = []
+    for rows, cols in [(3, 8), (17, 64), (511, 128), (1000, 8), (16, 2048)]:
+        big = torch.randn(rows, cols * 3, device="cuda")
+        src = big[:, cols : 2 * cols]  # strided, 16B-aligned rows (f32)
+        dst = torch.zeros(rows, cols, device="cuda")
+        pairs.append((src, dst))
+        expect.append(src.clone())
+    gpu.copy_pairs(pairs, torch.device("cuda", 0), blocking=True)
+    torch.cuda.synchronize()
+    for (src, dst), exp in zip(pairs, expect):
+        assert torch.equal(dst, exp)

@@ -251,7 +251,10 @@ struct SliceDesc {
   uint32_t row_bytes;
   uint32_t tiles_per_row;
   uint32_t ndim;            // number of outer dims
-  uint32_t _pad;
+  // >1 = row-packed: one unit covers this many consecutive rows (small
+  // rows, ndim<=1, fully 16B-aligned — the reshard scatter case, where a
+  // per-row unit would pay desc-load + search overhead per ~1 KB of work)
+  uint32_t rows_per_unit;
   uint64_t shape[kMaxDims];       // outer dims, innermost-last
   int64_t src_stride[kMaxDims];   // byte strides of outer dims
   int64_t dst_stride[kMaxDims];
@@ -294,17 +297,81 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
   }
   const uint32_t wave = threadIdx.x >> 6;
   const uint32_t lane = threadIdx.x & 63u;
-  const uint64_t stride = (uint64_t)gridDim.x * 4;
-  for (uint64_t unit = (uint64_t)blockIdx.x * 4 + wave; unit < total_units;
-       unit += stride) {
-    // binary search: greatest s with units_prefix <= unit
-    uint32_t lo = 0, hi = nslices - 1;
-    while (lo < hi) {
-      uint32_t mid = (lo + hi + 1) >> 1;
-      if (descs[mid].units_prefix <= unit) lo = mid; else hi = mid - 1;
+  // BLOCK-CHUNKED assignment: each wave owns a CONTIGUOUS unit range, so
+  // the slice descriptor is loaded once and advanced linearly — a strided
+  // grid walk would re-search and re-load ~176 B of desc per unit, which
+  // dominated small-row scatters (measured 249 GB/s in round 1)
+  const uint64_t nwaves = (uint64_t)gridDim.x * 4;
+  const uint64_t wave_id = (uint64_t)blockIdx.x * 4 + wave;
+  const uint64_t chunk = (total_units + nwaves - 1) / nwaves;
+  uint64_t unit = wave_id * chunk;
+  const uint64_t unit_end =
+      unit + chunk < total_units ? unit + chunk : total_units;
+  if (unit >= unit_end) return;
+  // binary search ONCE for the range start: greatest s with prefix <= unit
+  uint32_t lo = 0, hi = nslices - 1;
+  while (lo < hi) {
+    uint32_t mid = (lo + hi + 1) >> 1;
+    if (descs[mid].units_prefix <= unit) lo = mid; else hi = mid - 1;
+  }
+  SliceDesc d = descs[lo];
+  uint64_t next_prefix = (lo + 1 < nslices)
+                             ? descs[lo + 1].units_prefix
+                             : ~0ull;
+  for (; unit < unit_end; ++unit) {
+    while (unit >= next_prefix) {
+      ++lo;
+      d = descs[lo];
+      next_prefix = (lo + 1 < nslices) ? descs[lo + 1].units_prefix : ~0ull;
     }
-    const SliceDesc d = descs[lo];
     uint64_t local = unit - d.units_prefix;
+    if (d.rows_per_unit > 1) {
+      // row-packed unit: rows_per_unit consecutive rows, ndim<=1, all
+      // 16B-aligned; a flat vectorized loop with div/mod row recovery
+      // keeps ~4 independent 16B accesses in flight per lane
+      uint64_t row0 = local * d.rows_per_unit;
+      uint32_t nrows = (uint32_t)(d.rows - row0 < d.rows_per_unit
+                                      ? d.rows - row0
+                                      : d.rows_per_unit);
+      int64_t sstride = d.ndim ? d.src_stride[0] : d.row_bytes;
+      int64_t dstride = d.ndim ? d.dst_stride[0] : d.row_bytes;
+      const char* sbase =
+          reinterpret_cast<const char*>(d.src) + (int64_t)row0 * sstride;
+      char* dbase = reinterpret_cast<char*>(d.dst) + (int64_t)row0 * dstride;
+      const uint32_t rb4 = d.row_bytes >> 4;
+      const uint32_t elems = nrows * rb4;
+      uint32_t i = lane;
+      for (; i + 192 < elems; i += 256) {
+        uint32_t i0 = i, i1 = i + 64, i2 = i + 128, i3 = i + 192;
+        const uint4* s0 = reinterpret_cast<const uint4*>(
+            sbase + (int64_t)(i0 / rb4) * sstride + ((i0 % rb4) << 4));
+        const uint4* s1 = reinterpret_cast<const uint4*>(
+            sbase + (int64_t)(i1 / rb4) * sstride + ((i1 % rb4) << 4));
+        const uint4* s2 = reinterpret_cast<const uint4*>(
+            sbase + (int64_t)(i2 / rb4) * sstride + ((i2 % rb4) << 4));
+        const uint4* s3 = reinterpret_cast<const uint4*>(
+            sbase + (int64_t)(i3 / rb4) * sstride + ((i3 % rb4) << 4));
+        uint4 v0 = *s0;
+        uint4 v1 = *s1;
+        uint4 v2 = *s2;
+        uint4 v3 = *s3;
+        *reinterpret_cast<uint4*>(
+            dbase + (int64_t)(i0 / rb4) * dstride + ((i0 % rb4) << 4)) = v0;
+        *reinterpret_cast<uint4*>(
+            dbase + (int64_t)(i1 / rb4) * dstride + ((i1 % rb4) << 4)) = v1;
+        *reinterpret_cast<uint4*>(
+            dbase + (int64_t)(i2 / rb4) * dstride + ((i2 % rb4) << 4)) = v2;
+        *reinterpret_cast<uint4*>(
+            dbase + (int64_t)(i3 / rb4) * dstride + ((i3 % rb4) << 4)) = v3;
+      }
+      for (; i < elems; i += 64) {
+        *reinterpret_cast<uint4*>(
+            dbase + (int64_t)(i / rb4) * dstride + ((i % rb4) << 4)) =
+            *reinterpret_cast<const uint4*>(
+                sbase + (int64_t)(i / rb4) * sstride + ((i % rb4) << 4));
+      }
+      continue;
+    }
     uint64_t row = local / d.tiles_per_row;
     uint32_t tile = (uint32_t)(local % d.tiles_per_row);
     int64_t soff, doff;
@@ -534,8 +601,23 @@ static void copy_slices(const std::vector<PySlice>& slices, int device,
     d.row_bytes = (uint32_t)row_bytes;
     d.tiles_per_row = (uint32_t)((row_bytes + tile - 1) / tile);
     if (d.tiles_per_row == 0) d.tiles_per_row = 1;
+    d.rows_per_unit = 1;
+    // row-packed eligibility: small fully-16B-aligned rows, <=1 outer dim
+    bool aligned16 = row_bytes > 0 && (row_bytes & 15u) == 0 &&
+                     (d.src & 15u) == 0 && (d.dst & 15u) == 0 &&
+                     (d.ndim == 0 ||
+                      ((d.src_stride[0] & 15) == 0 &&
+                       (d.dst_stride[0] & 15) == 0));
+    if (d.ndim <= 1 && d.rows > 1 && row_bytes < tile && aligned16) {
+      d.rows_per_unit = (uint32_t)(tile / row_bytes);
+      d.tiles_per_row = 1;
+    }
     d.units_prefix = units;
-    units += d.rows * d.tiles_per_row;
+    if (d.rows_per_unit > 1) {
+      units += (d.rows + d.rows_per_unit - 1) / d.rows_per_unit;
+    } else {
+      units += d.rows * d.tiles_per_row;
+    }
   }
   if (units == 0) return;
 
