@@ -57,6 +57,23 @@ def main():
     dt = timeit("ipc_export x64 tensors", export_many, n=5)
     print(f"  per export: {dt/64*1e3:.3f} ms")
 
+    # COLD export cost (dmabuf ioctl) — cache cleared between calls
+    def export_cold():
+        e.ipc_export_cache_clear()
+        for x in ts[:8]:
+            e.ipc_export(x.data_ptr(), 0, 0)
+
+    dt = timeit("ipc_export x8 COLD (1MB blocks)", export_cold, n=3)
+    print(f"  per cold export: {dt/8*1e3:.3f} ms")
+    big_t = torch.empty(256 << 20, dtype=torch.uint8, device="cuda")
+
+    def export_cold_big():
+        e.ipc_export_cache_clear()
+        e.ipc_export(big_t.data_ptr(), 0, 0)
+
+    dt = timeit("ipc_export COLD 256MB block", export_cold_big, n=3)
+    e.ipc_export_cache_clear()
+
     # D2D copy bandwidth via copy_batch (same device)
     for size_mb in [1, 64, 1024]:
         n = size_mb * (1 << 20)
@@ -153,6 +170,17 @@ def main():
         torch.cuda.synchronize()
 
     dt = timeit("torch copy_ loop same scatter", torch_loop, n=10)
+    print(f"  bw (rd+wr): {nb2/dt/1e9:.1f} GB/s")
+
+    # SDMA ceiling for the same contig->strided 1KB/256KB write pattern
+    def sdma_scatter():
+        e.copy_batch_2d([
+            (dest_big.data_ptr() + i * 512 * 2, 0, 512 * n_slices * 2,
+             srcs2[i].data_ptr(), 0, 512 * 2, 512 * 2, 512)
+            for i in range(n_slices)
+        ])
+
+    dt = timeit("hipMemcpy2D x256 same scatter", sdma_scatter, n=10)
     print(f"  bw (rd+wr): {nb2/dt/1e9:.1f} GB/s")
 
     # K3 cast bandwidth

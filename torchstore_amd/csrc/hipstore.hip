@@ -233,14 +233,20 @@ static void copy_batch_2d(
 static constexpr int kMaxDims = 7;       // outer dims (innermost is the row)
 // tile size per launch: big batches amortize per-tile overhead with 128 KiB
 // tiles; smaller ones keep 16 KiB tiles so enough waves stay busy
-// (HIPSTORE_TILE overrides for experiments)
+// (HIPSTORE_TILE overrides for experiments; read per call so tuning
+// sweeps can vary it without a fresh process)
 static uint32_t pick_tile(uint64_t total_bytes) {
-  static uint32_t forced = [] {
-    const char* e = getenv("HIPSTORE_TILE");
-    return e ? (uint32_t)atoi(e) : 0u;
-  }();
+  const char* e = getenv("HIPSTORE_TILE");
+  uint32_t forced = e ? (uint32_t)atoi(e) : 0u;
   if (forced >= 4096) return forced;
   return total_bytes > (512ull << 20) ? 131072u : 16384u;
+}
+
+// grid cap for the slice kernel (blocks); HIPSTORE_GRID overrides
+static uint32_t pick_grid_cap() {
+  const char* e = getenv("HIPSTORE_GRID");
+  uint32_t forced = e ? (uint32_t)atoi(e) : 0u;
+  return forced >= 64 ? forced : 2048u;
 }
 
 struct SliceDesc {
@@ -449,11 +455,10 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
 }
 
 static bool use_nt_stores() {
-  static int v = [] {
-    const char* e = getenv("HIPSTORE_NT");
-    return e ? atoi(e) : 1;  // default decided by the hardware microbench
-  }();
-  return v != 0;
+  const char* e = getenv("HIPSTORE_NT");
+  // default ON: +5% on 1 GiB flat copies, +26% on the reshard scatter
+  // (profiles/gpu_primitive_microbench.log, NT1 vs NT0 runs)
+  return e ? atoi(e) != 0 : true;
 }
 
 // upload descriptors through the pinned staging buffer + launch the kernel
@@ -476,9 +481,9 @@ static void launch_slice_descs(std::vector<SliceDesc>& descs, uint64_t units,
   std::memcpy(p.h_desc, descs.data(), bytes);
   HIP_CHECK(hipMemcpyAsync(p.d_desc, p.h_desc, bytes, hipMemcpyHostToDevice,
                            stream));
-  // memory-bound: cap grid at 2048 blocks, grid-stride the rest (guide
-  // G11); each block consumes 4 units (one per wave)
-  uint32_t grid = (uint32_t)std::min<uint64_t>((units + 3) / 4, 2048);
+  // memory-bound: cap the grid and chunk units over it (guide G11);
+  // each block consumes 4 contiguous unit ranges (one per wave)
+  uint32_t grid = (uint32_t)std::min<uint64_t>((units + 3) / 4, pick_grid_cap());
   auto kern = use_nt_stores()
                   ? (remote ? copy_slices_kernel<true, true>
                             : copy_slices_kernel<true, false>)
@@ -691,7 +696,8 @@ template <typename SrcT, typename DstT>
 static void launch_cast(uintptr_t src, uintptr_t dst, uint64_t numel,
                         hipStream_t stream) {
   uint64_t work = (numel + 7) / 8;
-  uint32_t grid = (uint32_t)std::min<uint64_t>((work + 255) / 256, 2048);
+  uint32_t grid =
+      (uint32_t)std::min<uint64_t>((work + 255) / 256, pick_grid_cap());
   if (grid == 0) grid = 1;
   hipLaunchKernelGGL((cast_copy_kernel<SrcT, DstT>), dim3(grid), dim3(256), 0,
                      stream, reinterpret_cast<const SrcT*>(src),

@@ -236,7 +236,9 @@ class HipIpcTransportBuffer(TransportBuffer):
         return state
 
     # -- chunked windows (client side) -----------------------------------
-    async def _chunked_put_windows(self, t: torch.Tensor) -> str:
+    async def _chunked_put_windows(
+        self, t: torch.Tensor, request: Optional[Request] = None
+    ) -> str:
         """Stream a big tensor into volume staging, window by window.
 
         Double-buffered over two staging chunks: while window N's commit
@@ -245,8 +247,9 @@ class HipIpcTransportBuffer(TransportBuffer):
         volume = self._volume_ref.volume
         cache: IpcOpenCache = self._client_ctx.cache(IpcOpenCache)
         token = uuid.uuid4().hex
+        meta = request.meta_only() if request is not None else None
         staging_descs = await volume.handshake.call_one(
-            self, (token, tuple(t.shape), t.dtype), "chunk_put_init"
+            self, (token, meta, tuple(t.shape), t.dtype), "chunk_put_init"
         )
         try:
             ptrs = [
@@ -349,8 +352,24 @@ class HipIpcTransportBuffer(TransportBuffer):
         cache: ChunkStagingCache = self._volume_ctx.cache(ChunkStagingCache)
         device = volume.device
         if phase == "chunk_put_init":
-            token, shape, dtype = args
-            payload = torch.empty(shape, dtype=dtype, device=device)
+            token, meta, shape, dtype = args
+            payload = None
+            store = getattr(volume, "store", None)
+            if meta is not None and store is not None:
+                # re-put of an existing key: write INTO the stored tensor
+                # (the non-chunked path's `prior` reuse) — a fresh multi-GB
+                # hipMalloc per re-put cost ~30 ms and doubled peak memory
+                prior = store.find_existing(meta)
+                if (
+                    prior is not None
+                    and tuple(prior.shape) == tuple(shape)
+                    and prior.dtype == dtype
+                    and prior.is_contiguous()
+                    and prior.device == device
+                ):
+                    payload = prior
+            if payload is None:
+                payload = torch.empty(shape, dtype=dtype, device=device)
             return cache.acquire(token, payload, device)
         if phase == "chunk_put_commit":
             token, chunk_idx, dst_off, length = args
@@ -501,7 +520,7 @@ class HipIpcTransportBuffer(TransportBuffer):
         for i, tc in staged:
             desc = try_export(tc, gens.get(tc.device.index))
             if desc is None:
-                token = await self._chunked_put_windows(tc)
+                token = await self._chunked_put_windows(tc, requests[i])
                 payload[i] = ("chunked", token)
             else:
                 payload[i] = ("ipc", desc)
