@@ -623,7 +623,7 @@ class HipIpcTransportBuffer(TransportBuffer):
         return -1
 
     async def _stage_get_normal(
-        self, i: int, r: Request, synced: set
+        self, i: int, r: Request, synced: set, gens: Optional[Dict] = None
     ) -> Tuple[str, Any]:
         """Per-request direct staging: export the dest (or a dense scratch),
         falling back to the windowed path for >=2 GiB blocks."""
@@ -639,7 +639,18 @@ class HipIpcTransportBuffer(TransportBuffer):
             torch.cuda.current_stream(target.device).synchronize()
             synced.add(target.device.index)
         self._hold.append(target)
-        desc = try_export(target)
+        gen = None
+        if gens is not None:
+            di = target.device.index
+            gen = gens.get(di)
+            if gen is None:
+                # one memory_stats call per device per BATCH (it costs
+                # ~100 µs — per-tensor it was 30 ms on a Llama-8B dict)
+                from torchstore_amd.ops import gpu as gpu_ops
+
+                gen = gpu_ops.alloc_generation(di)
+                gens[di] = gen
+        desc = try_export(target, gen)
         if desc is None:
             token = await self._chunked_get_windows(r, target)
             return ("chunked", token)
@@ -648,6 +659,7 @@ class HipIpcTransportBuffer(TransportBuffer):
     async def client_stage_get(self, requests: Sequence[Request]) -> None:
         payload: List[Tuple[str, Any]] = []
         synced: set = set()
+        gens: Dict[int, int] = {}  # per-device allocator generation, per batch
         vol_dev = self._volume_device_index()
         # cross-device small/strided pieces coalesce through ONE bounce
         # buffer per op: the volume packs them locally (one kernel), moves
@@ -671,7 +683,7 @@ class HipIpcTransportBuffer(TransportBuffer):
                 payload.append(("bounce", None))  # offsets assigned below
                 bounce_plan.append((i, nbytes))
                 continue
-            payload.append(await self._stage_get_normal(i, r, synced))
+            payload.append(await self._stage_get_normal(i, r, synced, gens))
 
         self.bounce_descs: List[IpcDescriptor] = []
         self._bounces: List[torch.Tensor] = []
@@ -684,7 +696,7 @@ class HipIpcTransportBuffer(TransportBuffer):
             oversized = [e for e in aligned if e[2] > cap]
             aligned = [e for e in aligned if e[2] <= cap]
             for i, _n, _a in oversized:
-                payload[i] = await self._stage_get_normal(i, requests[i], synced)
+                payload[i] = await self._stage_get_normal(i, requests[i], synced, gens)
             remaining = sum(a for _, _, a in aligned)
             off = 0
             size = 0
@@ -713,7 +725,7 @@ class HipIpcTransportBuffer(TransportBuffer):
                 self.bounce_descs = []
                 for i, nbytes, a in aligned:
                     payload[i] = await self._stage_get_normal(
-                        i, requests[i], synced
+                        i, requests[i], synced, gens
                     )
             elif device.index not in synced:
                 torch.cuda.current_stream(device).synchronize()
