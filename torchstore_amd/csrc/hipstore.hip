@@ -235,18 +235,33 @@ static constexpr int kMaxDims = 7;       // outer dims (innermost is the row)
 // tiles; smaller ones keep 16 KiB tiles so enough waves stay busy
 // (HIPSTORE_TILE overrides for experiments; read per call so tuning
 // sweeps can vary it without a fresh process)
+// defaults from the hardware tile x grid sweep (profiles/kernel_tune.log):
+// 32 KiB tiles + an 8192-block grid cap won the flat bulk pattern (+5%
+// over 128 KiB/2048); small batches keep 16 KiB tiles
 static uint32_t pick_tile(uint64_t total_bytes) {
   const char* e = getenv("HIPSTORE_TILE");
   uint32_t forced = e ? (uint32_t)atoi(e) : 0u;
   if (forced >= 4096) return forced;
-  return total_bytes > (512ull << 20) ? 131072u : 16384u;
+  return total_bytes > (512ull << 20) ? 32768u : 16384u;
 }
 
 // grid cap for the slice kernel (blocks); HIPSTORE_GRID overrides
 static uint32_t pick_grid_cap() {
   const char* e = getenv("HIPSTORE_GRID");
   uint32_t forced = e ? (uint32_t)atoi(e) : 0u;
-  return forced >= 64 ? forced : 2048u;
+  return forced >= 64 ? forced : 8192u;
+}
+
+// row-packed units span this many bytes of small rows regardless of the
+// launch tile (64 KiB won the scatter sweep at every grid size)
+static constexpr uint32_t kRowPackSpan = 65536u;
+
+// the cast kernel prefers a SMALLER grid (1024 beat 2048 by ~8% in the
+// sweep — fewer blocks, longer per-block streams)
+static uint32_t pick_cast_grid_cap() {
+  const char* e = getenv("HIPSTORE_CAST_GRID");
+  uint32_t forced = e ? (uint32_t)atoi(e) : 0u;
+  return forced >= 64 ? forced : 1024u;
 }
 
 struct SliceDesc {
@@ -613,8 +628,8 @@ static void copy_slices(const std::vector<PySlice>& slices, int device,
                      (d.ndim == 0 ||
                       ((d.src_stride[0] & 15) == 0 &&
                        (d.dst_stride[0] & 15) == 0));
-    if (d.ndim <= 1 && d.rows > 1 && row_bytes < tile && aligned16) {
-      d.rows_per_unit = (uint32_t)(tile / row_bytes);
+    if (d.ndim <= 1 && d.rows > 1 && row_bytes < kRowPackSpan && aligned16) {
+      d.rows_per_unit = (uint32_t)(kRowPackSpan / row_bytes);
       d.tiles_per_row = 1;
     }
     d.units_prefix = units;
@@ -697,7 +712,7 @@ static void launch_cast(uintptr_t src, uintptr_t dst, uint64_t numel,
                         hipStream_t stream) {
   uint64_t work = (numel + 7) / 8;
   uint32_t grid =
-      (uint32_t)std::min<uint64_t>((work + 255) / 256, pick_grid_cap());
+      (uint32_t)std::min<uint64_t>((work + 255) / 256, pick_cast_grid_cap());
   if (grid == 0) grid = 1;
   hipLaunchKernelGGL((cast_copy_kernel<SrcT, DstT>), dim3(grid), dim3(256), 0,
                      stream, reinterpret_cast<const SrcT*>(src),
