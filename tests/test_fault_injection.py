@@ -118,8 +118,8 @@ async def _drive_put(buffer, t):
 
 
 def test_chunked_put_roundtrip_pipelined(chunked_env):
-    """Happy path: 5 windows double-buffered over 2 staging chunks land
-    bit-exact in the volume payload; the op's chunks return to the pool."""
+    """Happy path: windows pipelined over 3 staging chunks land bit-exact
+    in the volume payload; the op's chunks return to the pool."""
     buffer, volume, volume_ctx = chunked_env
     t = torch.arange(1200, dtype=torch.uint8)  # 2 full windows + partial
     token = asyncio.run(_drive_put(buffer, t))
@@ -128,7 +128,7 @@ def test_chunked_put_roundtrip_pipelined(chunked_env):
     assert torch.equal(payload, t)
     released = cache.release(token)
     assert released is payload
-    assert len(cache.free) == 2 and not cache.by_token
+    assert len(cache.free) == 3 and not cache.by_token
 
 
 @pytest.mark.parametrize("fail_phase,fail_idx", [
@@ -146,7 +146,7 @@ def test_chunked_put_abort_releases_staging(chunked_env, fail_phase, fail_idx):
     # the abort path's chunk_release must return every chunk to the pool
     assert not cache.by_token, "aborted op leaked staging chunks"
     if fail_phase != "chunk_put_init":
-        assert len(cache.free) == 2
+        assert len(cache.free) == 3
 
 
 @pytest.mark.parametrize("fail_phase,fail_idx", [
