@@ -231,3 +231,36 @@ def test_pg_pair_cache_publish_on_success(monkeypatch):
     obj_req = Request(key="o", objects={"x": 1}, is_object=True)
     asyncio.run(buffer2.put([obj_req]))
     assert cache.pairs["v0"].confirmed
+
+
+def test_pg_pair_retry_builds_fresh_rendezvous(monkeypatch):
+    """After a failed (discarded) pair, the next op creates a NEW pair id
+    and TCPStore — the failed rendezvous is never retried in place."""
+    from torchstore_amd.transport.pg import GlooTransportBuffer, PgClientCache
+
+    ctx = TransportContext()
+
+    class Ref:
+        volume_id = "v1"
+        volume = None
+
+    cache: PgClientCache = ctx.cache(PgClientCache)
+
+    b1 = GlooTransportBuffer()
+    b1.bind_client(Ref(), ctx)
+    first = cache.get_or_create("v1", "gloo")
+    first_id = first.info.pair_id
+
+    class Dead:
+        class put:  # noqa: N801
+            @staticmethod
+            async def call_one(*a, **kw):
+                raise ConnectionError("down")
+
+    b1._volume_ref.volume = Dead()
+    with pytest.raises(ConnectionError):
+        asyncio.run(b1.put([Request(key="o", objects=1, is_object=True)]))
+    assert "v1" not in cache.pairs
+    fresh = cache.get_or_create("v1", "gloo")
+    assert fresh.info.pair_id != first_id
+    assert fresh.info.port != 0
