@@ -86,6 +86,19 @@ def main():
         )
         print(f"  bw: {n/dt/1e9:.1f} GB/s")
 
+    # torch's own flat 1 GiB d2d copy — the honest apples-to-apples
+    # ceiling for copy_batch's kernel path (the 6.6 TB/s .contiguous
+    # number is a 128 MB working set resident in the 256 MB LLC)
+    a_t = torch.empty(1 << 30, dtype=torch.uint8, device="cuda")
+    b_t = torch.empty(1 << 30, dtype=torch.uint8, device="cuda")
+
+    def torch_flat():
+        b_t.copy_(a_t)
+        torch.cuda.synchronize()
+
+    dt = timeit("torch copy_ flat 1GiB d2d", torch_flat, n=10)
+    print(f"  bw: {(1 << 30)/dt/1e9:.1f} GB/s")
+
     # large-copy strategies: kernel vs SDMA variants
     n1g = 1 << 30
     a1 = torch.empty(n1g, dtype=torch.uint8, device="cuda")
