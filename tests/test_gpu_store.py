@@ -576,3 +576,91 @@ async def test_rccl_transport_on_hardware():
         assert torch.equal(got, t2)
 
     await _with_store(body, transport=TransportType.RCCL)
+
+
+class MultiRankWeightSource(Actor):
+    """Trainer stand-in registering EIGHT ranks' fsdp shards from one
+    process — the direct-sync N=8 source topology on one GPU."""
+
+    def __init__(self, controller, layers):
+        import os
+
+        torch.cuda.set_device(0)
+        ts.attach(controller, SingletonStrategy())
+        from torchstore_amd.models import llama
+
+        self.world = 8
+        self.sources = []
+        self.sds = []
+        for r in range(self.world):
+            os.environ["RANK"] = str(r)
+            sd = llama.make_local_shard_state_dict(
+                r, self.world, llama.fsdp_placement, device="cuda:0",
+                layers=layers, pattern=True,
+            )
+            self.sds.append(sd)
+
+    @endpoint
+    async def push_all(self):
+        from torchstore_amd.weight_sync import DirectWeightSyncSource
+
+        for r, sd in enumerate(self.sds):
+            src = DirectWeightSyncSource(
+                ts.client(), "d8", rank=r, world_size=self.world
+            )
+            await src.push(sd)
+            self.sources.append(src)
+        torch.cuda.synchronize()
+        return "ok"
+
+
+@requires_gpu
+async def test_direct_sync_fake8(monkeypatch):
+    """8 fsdp source shards → one tp dest rank, pulled one-sided: the
+    plan spans every source rank (pitched '2d' ops dominate) and executes
+    as ONE batched remote-read kernel launch — the N=8 direct-sync shape
+    on a single GPU."""
+    from torchstore_amd.models import llama
+    from torchstore_amd.types import LocalShard
+    from torchstore_amd.weight_sync import DirectWeightSyncDest
+
+    layers = 2
+    controller = await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(),
+        storage_device="auto",
+    )
+    mesh = None
+    try:
+        import asyncio as aio
+
+        from torchstore_amd.runtime import spawn_actors
+
+        mesh = await aio.to_thread(
+            spawn_actors, 1, MultiRankWeightSource, "w8", controller, layers
+        )
+        assert await mesh.handles[0].push_all.call_one() == "ok"
+        for rank in (0, 5):
+            monkeypatch.setenv("RANK", str(rank))
+            dst = llama.make_local_shard_state_dict(
+                rank, 8, llama.tp_placement, device="cuda:0", layers=layers,
+            )
+            dest = DirectWeightSyncDest(ts.client(), "d8")
+            await dest.pull(dst)
+            torch.cuda.synchronize()
+            kinds = [op.kind for op in dest._plan]
+            assert "2d" in kinds, "expected pitched multi-source reads"
+            for name, v in dst.items():
+                local = v.tensor if isinstance(v, LocalShard) else v
+                offsets = (
+                    v.slice.offsets if isinstance(v, LocalShard)
+                    else (0,) * local.dim()
+                )
+                exp = llama.expected_pattern(
+                    tuple(local.shape), offsets, local.dtype, local.device
+                )
+                assert torch.equal(local, exp), f"rank {rank} {name}"
+    finally:
+        if mesh is not None:
+            await mesh.stop()
+        await ts.shutdown()

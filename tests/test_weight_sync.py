@@ -241,3 +241,65 @@ async def test_missing_source_raises(fake_codec):
             await dest.pull({"w": torch.zeros(2)})
 
     await _with_store(body)
+
+
+async def test_coverage_gap_raises(fake_codec):
+    """Source shards that do NOT tile the wanted region must raise instead
+    of silently leaving stale weights (verify coverage)."""
+    from torchstore_amd.types import LocalShard
+
+    async def body():
+        c = ts.client()
+        # source registers only rows 0..32 of a 64-row parameter
+        top = torch.randn(32, 16)
+        src = DirectWeightSyncSource(c, "gap", rank=0, world_size=1)
+        await src.push({
+            "w": LocalShard(
+                tensor=top,
+                slice=TensorSlice(
+                    offsets=(0, 0), local_shape=(32, 16),
+                    global_shape=(64, 16), coordinates=(0,), mesh_shape=(2,),
+                ),
+            )
+        })
+        dest = DirectWeightSyncDest(c, "gap")
+        full = {"w": torch.zeros(64, 16)}
+        with pytest.raises(RuntimeError, match="do not tile"):
+            await dest.pull(full)
+        # a dest asking only for the covered half succeeds
+        half = {
+            "w": LocalShard(
+                tensor=torch.zeros(32, 16),
+                slice=TensorSlice(
+                    offsets=(0, 0), local_shape=(32, 16),
+                    global_shape=(64, 16), coordinates=(0,), mesh_shape=(2,),
+                ),
+            )
+        }
+        await dest.pull(half)
+        assert torch.equal(half["w"].tensor, top)
+
+    await _with_store(body)
+
+
+async def test_plan_cache_dest_identity(fake_codec):
+    """The cached plan is keyed by IDENTITY of the dest dict: a different
+    dict (even with equal shapes) forces a rebuild into the new tensors."""
+
+    async def body():
+        c = ts.client()
+        w = torch.randn(16, 16)
+        src = DirectWeightSyncSource(c, "ident", rank=0, world_size=1)
+        await src.push({"w": w})
+        dest = DirectWeightSyncDest(c, "ident")
+        d1 = {"w": torch.zeros(16, 16)}
+        await dest.pull(d1)
+        assert torch.equal(d1["w"], w)
+        n_reads = fake_codec.read_count
+        await dest.pull(d1)  # same dict -> cached plan
+        d2 = {"w": torch.zeros(16, 16)}
+        await dest.pull(d2)  # new dict -> rebuilt plan targets d2's tensor
+        assert torch.equal(d2["w"], w)
+        assert fake_codec.read_count == n_reads + 2
+
+    await _with_store(body)
