@@ -152,3 +152,40 @@ def test_shm_get_segments_are_client_keyed():
     d1b = asyncio.run(fetch("client-A"))
     assert d1b.seg_key == d1.seg_key
     volume_ctx.close()
+
+
+def test_shm_put_segment_reuse_and_growth():
+    """Re-put of the same key reuses the volume-owned segment; a LARGER
+    re-put allocates a bigger one (reference SHM descriptor-reuse
+    handshake, shared_memory.py:340-360)."""
+    import asyncio
+
+    import torch
+
+    from torchstore_amd.transport.base import TransportContext
+    from torchstore_amd.transport.shm import ShmTransportBuffer, ShmVolumeCache
+    from torchstore_amd.types import Request
+
+    volume_ctx = TransportContext()
+
+    def handshake(nbytes):
+        buf = ShmTransportBuffer()
+        buf.client_uid = "c1"
+        buf.alloc_sizes = [nbytes]
+        buf.attach_volume(volume_ctx)
+        (desc,) = buf.recv_handshake([Request(key="k")], "put", None)
+        return desc
+
+    d1 = handshake(1024)
+    d2 = handshake(512)       # smaller fits -> same segment
+    assert d2.seg_key == d1.seg_key
+    d3 = handshake(4096)      # larger -> fresh, bigger segment
+    assert d3.seg_key != d1.seg_key and d3.nbytes >= 4096
+    # a different client never shares the put segment
+    buf = ShmTransportBuffer()
+    buf.client_uid = "c2"
+    buf.alloc_sizes = [1024]
+    buf.attach_volume(volume_ctx)
+    (d4,) = buf.recv_handshake([Request(key="k")], "put", None)
+    assert d4.seg_key != d3.seg_key
+    volume_ctx.close()
