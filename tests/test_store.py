@@ -283,3 +283,23 @@ async def test_get_batch_plan_cache():
             await c.get_batch(fetches)
 
     await _with_store(TransportType.RPC, body)
+
+
+async def test_keys_dotted_prefix_api():
+    """API-level dotted-prefix filtering: keys('sd/model') matches dotted
+    module paths by component (reference StringTrie('.') semantics)."""
+
+    async def body():
+        await ts.put_state_dict(
+            {"model": {"layers": {"0": torch.ones(2), "1": torch.ones(2)}},
+             "opt": {"lr": 0.1}},
+            "sd",
+        )
+        ks = await ts.keys("sd/model.layers")
+        assert sorted(ks) == ["sd/model.layers.0", "sd/model.layers.1"]
+        ks2 = await ts.keys("sd/model")
+        assert sorted(ks2) == ["sd/model.layers.0", "sd/model.layers.1"]
+        # component boundary: "sd/mod" matches nothing
+        assert await ts.keys("sd/mod") == []
+
+    await _with_store(TransportType.RPC, body)
