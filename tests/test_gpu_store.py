@@ -664,3 +664,32 @@ async def test_direct_sync_fake8(monkeypatch):
         if mesh is not None:
             await mesh.stop()
         await ts.shutdown()
+
+
+@requires_gpu
+async def test_tiered_store_hbm_spill_to_host():
+    """HBM-primary tiered volume with a tiny capacity: overflow tensors
+    spill to host memory but serve back onto GPU dests transparently."""
+    await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(),
+        storage_device="auto",
+        storage_capacity_gb=0.01,  # 10 MB primary
+    )
+    try:
+        a = torch.randn(1024, 1024, device="cuda")  # 4 MB -> primary
+        b = torch.randn(1024, 1024, device="cuda")  # 4 MB -> primary
+        c = torch.randn(2048, 2048, device="cuda")  # 16 MB -> spills
+        await ts.put_batch({"t/a": a, "t/b": b, "t/c": c})
+        for k, v in (("t/a", a), ("t/b", b), ("t/c", c)):
+            dest = torch.zeros_like(v)
+            await ts.get(k, dest)
+            torch.cuda.synchronize()
+            assert torch.equal(dest, v), k
+        # slice fetch from the SPILLED tensor
+        from torchstore_amd.types import TensorSlice
+
+        got = await ts.get("t/c")
+        assert torch.equal(got, c)
+    finally:
+        await ts.shutdown()

@@ -50,6 +50,7 @@ async def initialize(
     strategy: Optional[PlacementStrategy] = None,
     store_name: str = DEFAULT_STORE,
     storage_device: str = "auto",
+    storage_capacity_gb: Optional[float] = None,
 ) -> ActorHandle:
     """Spawn volumes + controller; returns the controller handle.
 
@@ -71,6 +72,7 @@ async def initialize(
             f"{store_name}-volume",
             volume_id_seed=strategy.volume_id_seed,
             device=storage_device,
+            capacity_gb=storage_capacity_gb,
             # a node bringing up many ranks + volumes at once can take a
             # while per torch import; don't flake on a loaded box
             timeout=240.0,

@@ -96,22 +96,45 @@ class InMemoryStore(StorageImpl):
         self.device = torch.device(device)
         self.kv: Dict[str, Union[_ObjEntry, _TensorEntry, _ShardEntry]] = {}
 
+    # -- placement hooks (overridden by TieredStore) ----------------------
+    def _place(self, tensor: torch.Tensor) -> torch.Tensor:
+        """Decide where a stored tensor's bytes live."""
+        if tensor.device != self.device:
+            tensor = tensor.to(self.device)
+        return tensor
+
+    def _release(self, tensor: torch.Tensor) -> None:
+        """Accounting hook: ``tensor`` is leaving the store."""
+        return None
+
+    def _release_entry(self, entry) -> None:
+        if isinstance(entry, _TensorEntry):
+            self._release(entry.tensor)
+        elif isinstance(entry, _ShardEntry):
+            for _s, t in entry.shards.values():
+                self._release(t)
+
     # -- write -----------------------------------------------------------
     def put(self, request: Request, value: Any) -> None:
         key = request.key
+        old = self.kv.get(key)
         if request.is_object:
+            if old is not None:
+                self._release_entry(old)
             self.kv[key] = _ObjEntry(value)
             return
         tensor = value
         if not isinstance(tensor, torch.Tensor):
             raise TypeError(f"put of non-tensor {type(tensor)} without is_object")
-        if tensor.device != self.device:
-            tensor = tensor.to(self.device)
         if request.tensor_slice is None:
-            self.kv[key] = _TensorEntry(tensor)
+            if old is not None:
+                self._release_entry(old)
+            self.kv[key] = _TensorEntry(self._place(tensor))
             return
-        entry = self.kv.get(key)
-        if not isinstance(entry, _ShardEntry):
+        entry = old if isinstance(old, _ShardEntry) else None
+        if entry is None:
+            if old is not None:
+                self._release_entry(old)
             entry = _ShardEntry()
             self.kv[key] = entry
         ts = request.tensor_slice
@@ -123,8 +146,13 @@ class InMemoryStore(StorageImpl):
             ):
                 # a re-push under a different sharding: stale shards of the
                 # old layout could otherwise serve wrong regions
+                for _s, t in entry.shards.values():
+                    self._release(t)
                 entry.shards.clear()
-        entry.shards[ts.coordinates] = (ts, tensor)
+        prev = entry.shards.get(ts.coordinates)
+        if prev is not None:
+            self._release(prev[1])
+        entry.shards[ts.coordinates] = (ts, self._place(tensor))
 
     def find_existing(self, request: Request) -> Optional[torch.Tensor]:
         entry = self.kv.get(request.key)
@@ -188,8 +216,9 @@ class InMemoryStore(StorageImpl):
 
     # -- admin -----------------------------------------------------------
     def delete(self, key: str, missing_ok: bool = False) -> None:
-        if key in self.kv:
-            del self.kv[key]
+        entry = self.kv.pop(key, None)
+        if entry is not None:
+            self._release_entry(entry)
         elif not missing_ok:
             raise KeyError(key)
 
@@ -197,7 +226,65 @@ class InMemoryStore(StorageImpl):
         return list(self.kv.keys())
 
     def reset(self) -> None:
+        for entry in self.kv.values():
+            self._release_entry(entry)
         self.kv.clear()
+
+
+class TieredStore(InMemoryStore):
+    """HBM-primary store with a host-memory overflow tier.
+
+    Fills the ``StorageImpl`` seam the reference leaves open
+    (torchstore ``storage_volume.py:102-143``: "a second backend, e.g.
+    GPU-resident or tiered, is an explicit extension point"): tensors
+    live in the volume's HBM until ``capacity_bytes`` of primary
+    residency, then overflow to ``spill_device`` (pageable host memory).
+    Fetches/slices serve from either tier transparently — transports
+    route CPU-resident results over the inline/SHM paths.  Placement is
+    decided at write time; freeing primary bytes (delete/overwrite/
+    reset) makes room for later puts.  No eviction of already-resident
+    entries — the store is not a cache, readers hold zero-copy views.
+    """
+
+    def __init__(
+        self,
+        device: Union[str, torch.device],
+        capacity_bytes: int,
+        spill_device: Union[str, torch.device] = "cpu",
+    ):
+        super().__init__(device)
+        self.capacity_bytes = int(capacity_bytes)
+        self.spill_device = torch.device(spill_device)
+        self.primary_used = 0
+        self._resident: set = set()  # id() of primary-resident tensors
+
+    def _place(self, tensor: torch.Tensor) -> torch.Tensor:
+        nb = tensor.numel() * tensor.element_size()
+        if self.primary_used + nb <= self.capacity_bytes:
+            if tensor.device != self.device:
+                tensor = tensor.to(self.device)
+            self.primary_used += nb
+            self._resident.add(id(tensor))
+            return tensor
+        logger.info(
+            "tiered store: spilling %d bytes of %r to %s "
+            "(primary %d/%d used)",
+            nb, tensor.shape, self.spill_device, self.primary_used,
+            self.capacity_bytes,
+        )
+        if tensor.device != self.spill_device:
+            tensor = tensor.to(self.spill_device)
+        return tensor
+
+    def _release(self, tensor: torch.Tensor) -> None:
+        if id(tensor) in self._resident:
+            self._resident.discard(id(tensor))
+            self.primary_used -= tensor.numel() * tensor.element_size()
+
+    def reset(self) -> None:
+        super().reset()
+        self.primary_used = 0
+        self._resident.clear()
 
 
 def _resolve_device(device: str) -> torch.device:
@@ -214,12 +301,23 @@ def _resolve_device(device: str) -> torch.device:
 class StorageVolume(Actor):
     """A storage actor process; holds one StorageImpl + a transport context."""
 
-    def __init__(self, volume_id_seed: str = "rank", device: str = "auto"):
+    def __init__(
+        self,
+        volume_id_seed: str = "rank",
+        device: str = "auto",
+        capacity_gb: Optional[float] = None,
+        spill_device: str = "cpu",
+    ):
         self.ctx = TransportContext()
         self.device = _resolve_device(device)
         if self.device.type == "cuda":
             torch.cuda.set_device(self.device)
-        self.store = InMemoryStore(self.device)
+        if capacity_gb is not None:
+            self.store: StorageImpl = TieredStore(
+                self.device, int(capacity_gb * 1e9), spill_device
+            )
+        else:
+            self.store = InMemoryStore(self.device)
         self.volume_id = self._make_volume_id(volume_id_seed)
         self.hostname = os.environ.get("HOSTNAME") or socket.gethostname()
         logger.info(
