@@ -116,6 +116,7 @@ async def initialize_spmd(
     env: Optional[SPMDEnv] = None,
     store_name: str = api.DEFAULT_STORE,
     storage_device: str = "auto",
+    storage_capacity_gb: float = None,
     timeout_s: float = 300.0,
 ) -> ActorHandle:
     """Collective store bring-up across a torchrun world.
@@ -155,6 +156,7 @@ async def initialize_spmd(
             strategy=strategy,
             store_name=store_name,
             storage_device=storage_device,
+            storage_capacity_gb=storage_capacity_gb,
         )
         store.set(key, pickle.dumps(controller))
         store.add(count_key, per_host)
@@ -164,7 +166,7 @@ async def initialize_spmd(
         if env.local_rank == 0 and host_index > 0:
             local_mesh, infos = await _spawn_host_volumes(
                 strategy, store_name, storage_device, env, host_index,
-                per_host,
+                per_host, storage_capacity_gb,
             )
             await controller.register_volumes.call_one(infos)
             store.add(count_key, len(infos))
@@ -189,6 +191,7 @@ async def _spawn_host_volumes(
     env: SPMDEnv,
     host_index: int,
     per_host: int,
+    storage_capacity_gb: float = None,
 ):
     """Spawn this host's volume processes (called on local-rank-0 of every
     non-zero host) and build their registration infos."""
@@ -207,6 +210,7 @@ async def _spawn_host_volumes(
         f"{store_name}-volume-h{host_index}",
         volume_id_seed=seed,
         device=storage_device,
+        capacity_gb=storage_capacity_gb,
         timeout=240.0,
     )
     ids = await mesh.get_id.call()
