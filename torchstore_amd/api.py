@@ -78,7 +78,10 @@ async def initialize(
             timeout=240.0,
         ),
         asyncio.to_thread(
-            spawn_actor, Controller, f"{store_name}-controller", store_name
+            spawn_actor, Controller, f"{store_name}-controller", store_name,
+            # same generous timeout as the volumes: a loaded box paging
+            # torch into a fresh process can take >60 s
+            timeout=240.0,
         ),
     )
     ids = await mesh.get_id.call()

@@ -87,7 +87,9 @@ def spawn_actors(
     *args,
     mesh_shape: Optional[Sequence[int]] = None,
     env_per_rank: Optional[Dict[int, Dict[str, str]]] = None,
-    timeout: float = 60.0,
+    # a loaded box paging torch into a fresh spawn-context process can
+    # take >60 s (measured mid-suite on the GPU pool) — be generous
+    timeout: float = 240.0,
     **kwargs,
 ) -> ActorMesh:
     """Spawn ``num`` actor processes and return their mesh (blocking)."""
