@@ -303,3 +303,47 @@ async def test_keys_dotted_prefix_api():
         assert await ts.keys("sd/mod") == []
 
     await _with_store(TransportType.RPC, body)
+
+
+async def test_reput_across_volumes_serves_newest(monkeypatch):
+    """A re-put routed to a DIFFERENT volume (client identity changed)
+    must win over the stale copy on the old volume — the controller's
+    write sequence decides, never locality/lexical order."""
+    from torchstore_amd.strategy import LocalRankStrategy
+
+    await ts.initialize(
+        num_storage_volumes=2,
+        strategy=LocalRankStrategy(),
+        storage_device="cpu",
+    )
+    try:
+        monkeypatch.setenv("RANK", "0")
+        await ts.put("x", torch.zeros(32))          # volume 0
+        monkeypatch.setenv("RANK", "1")
+        await ts.put("x", torch.ones(32))           # volume 1 (newer)
+        for r in ("0", "1"):
+            monkeypatch.setenv("RANK", r)
+            out = await ts.get("x")
+            assert out.eq(1.0).all(), f"stale copy served for reader rank {r}"
+        # objects too
+        monkeypatch.setenv("RANK", "1")
+        await ts.put("o", {"v": 1})
+        monkeypatch.setenv("RANK", "0")
+        await ts.put("o", {"v": 2})
+        assert (await ts.get("o"))["v"] == 2
+        # keys()/exists see the newest kind: overwrite a half-committed
+        # shard set with a plain tensor -> readable immediately
+        from torchstore_amd.types import LocalShard, TensorSlice
+
+        monkeypatch.setenv("RANK", "0")
+        await ts.put("h", LocalShard(
+            tensor=torch.zeros(8, 4),
+            slice=TensorSlice((0, 0), (8, 4), (16, 4), (0,), (2,)),
+        ))  # half-committed: coordinate (1,) missing
+        assert not await ts.exists("h")
+        monkeypatch.setenv("RANK", "1")
+        await ts.put("h", torch.full((16, 4), 7.0))  # plain overwrite
+        assert await ts.exists("h")
+        assert (await ts.get("h")).eq(7.0).all()
+    finally:
+        await ts.shutdown()

@@ -211,22 +211,24 @@ class LocalClient:
         self, key: str, like: Any, locations: Dict[str, StorageInfo]
     ) -> List[Tuple[str, _SubFetch]]:
         """Decide which volumes to hit and what to ask each for."""
-        kinds = {info.object_type for info in locations.values()}
+        # the NEWEST write decides both the key's kind and — for whole
+        # tensors/objects — which volume's copy to serve: a re-put routed
+        # to a different volume (strategy/client change) leaves the old
+        # copy behind, and locality order must never resurrect it
+        newest_vid = max(locations, key=lambda v: locations[v].seq)
+        newest = locations[newest_vid]
         volume_ids = self._order_by_locality(locations.keys())
 
-        if kinds == {ObjectType.OBJECT}:
+        if newest.object_type == ObjectType.OBJECT:
             return [
-                (volume_ids[0], _SubFetch(key, Request(key=key, is_object=True)))
+                (newest_vid, _SubFetch(key, Request(key=key, is_object=True)))
             ]
 
         dest, wanted = self._dest_and_region(like)
 
-        if ObjectType.TENSOR in kinds:
-            # at least one volume has the whole tensor — single-volume fetch
-            vid = next(
-                v for v in volume_ids
-                if locations[v].object_type == ObjectType.TENSOR
-            )
+        if newest.object_type == ObjectType.TENSOR:
+            # the newest whole-tensor copy — single-volume fetch
+            vid = newest_vid
             req = Request(key=key, tensor_slice=wanted, tensor_val=dest)
             if dest is not None:
                 req.inplace = True
@@ -239,10 +241,10 @@ class LocalClient:
 
         # sharded key: expand into per-stored-shard intersections
         if wanted is None:
-            # full-tensor fetch of a sharded key: region = whole global shape
-            any_slice = next(
-                iter(next(iter(locations.values())).tensor_slices)
-            )
+            # full-tensor fetch of a sharded key: region = whole global
+            # shape (from the newest entry — a stale whole-tensor location
+            # may coexist with no slices at all)
+            any_slice = next(iter(newest.tensor_slices))
             wanted = _full_region_slice(any_slice.global_shape)
 
         subs: List[Tuple[str, _SubFetch]] = []

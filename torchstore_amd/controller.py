@@ -37,8 +37,14 @@ class ObjectType(Enum):
 class StorageInfo:
     object_type: ObjectType
     tensor_slices: Set[TensorSlice] = field(default_factory=set)
+    # controller-assigned write sequence: readers of a key stored on
+    # several volumes (e.g. a re-put routed to a different volume after a
+    # strategy/client change) must serve the NEWEST copy, not whichever
+    # volume sorts first — the stale-copy gap the reference leaves open
+    seq: int = 0
 
     def merge(self, other: "StorageInfo") -> None:
+        self.seq = max(self.seq, other.seq)
         if other.object_type != self.object_type:
             # last write wins on type change (e.g. object overwritten by tensor)
             self.object_type = other.object_type
@@ -99,6 +105,7 @@ class Controller(Actor):
         self.index: Trie = Trie()
         self.volumes: Dict[str, VolumeInfo] = {}
         self.strategy_spec: Optional[dict] = None
+        self._write_seq = 0
 
     # -- bring-up ---------------------------------------------------------
     @endpoint
@@ -120,11 +127,16 @@ class Controller(Actor):
     # -- commit tracking --------------------------------------------------
     @staticmethod
     def _is_fully_committed(locations: Dict[str, StorageInfo]) -> bool:
+        # the NEWEST write decides the key's current kind: a stale plain-
+        # tensor copy left by an earlier epoch must not make a half-
+        # committed shard set readable (and vice versa)
+        newest = max(locations.values(), key=lambda i: i.seq)
+        if newest.object_type != ObjectType.TENSOR_SLICE:
+            return True  # whole tensors/objects commit atomically
         slices: Set[TensorSlice] = set()
         for info in locations.values():
-            if info.object_type != ObjectType.TENSOR_SLICE:
-                return True  # whole tensors/objects commit atomically
-            slices |= info.tensor_slices
+            if info.object_type == ObjectType.TENSOR_SLICE:
+                slices |= info.tensor_slices
         if not slices:
             return False
         mesh_shape = next(iter(slices)).mesh_shape
@@ -144,6 +156,8 @@ class Controller(Actor):
                     "controller must only receive meta-only requests"
                 )
             info = _info_from_request(r)
+            self._write_seq += 1
+            info.seq = self._write_seq
             locations: Dict[str, StorageInfo] = self.index.get(r.key)
             if locations is None:
                 locations = {}
