@@ -91,3 +91,120 @@ def test_llama8b_total_size():
     )
     # Meta-Llama-3-8B has 8.03B parameters
     assert 8.0e9 < n_params < 8.1e9, n_params
+
+
+# ---------------------------------------------------------------------------
+# FSDP2 (fully_shard) real-model exchange across independent worlds —
+# the reference's workhorse (tests/test_state_dict.py:82-161, test_models.py)
+# ---------------------------------------------------------------------------
+
+import asyncio  # noqa: E402
+import os  # noqa: E402
+import tempfile  # noqa: E402
+import uuid  # noqa: E402
+
+from torchstore_amd.runtime import (  # noqa: E402
+    Actor,
+    actor_context,
+    close_connections,
+    endpoint,
+    spawn_actors,
+)
+
+
+class FsdpWorker(Actor):
+    """One rank of a fully_shard world sharing a tiny transformers model."""
+
+    def __init__(self, world, pg_file, controller, seed):
+        import torch.distributed as dist
+
+        self.rank = actor_context().rank
+        self.world = world
+        os.environ["RANK"] = str(self.rank)
+        dist.init_process_group(
+            "gloo", init_method=f"file://{pg_file}",
+            rank=self.rank, world_size=world,
+        )
+        from torchstore_amd import api
+        from torchstore_amd.strategy import LocalRankStrategy
+
+        api.attach(controller, LocalRankStrategy())
+        from torch.distributed.device_mesh import init_device_mesh
+
+        self.mesh = init_device_mesh("cpu", (world,))
+        self.seed = seed
+
+    def _model(self):
+        import transformers
+
+        cfg = transformers.LlamaConfig(
+            hidden_size=64, intermediate_size=128, num_attention_heads=4,
+            num_key_value_heads=2, num_hidden_layers=2, vocab_size=256,
+        )
+        torch.manual_seed(self.seed)
+        model = transformers.LlamaForCausalLM(cfg)
+        from torch.distributed.fsdp import fully_shard
+
+        for layer in model.model.layers:
+            fully_shard(layer, mesh=self.mesh)
+        fully_shard(model, mesh=self.mesh)
+        return model
+
+    @endpoint
+    async def push(self):
+        model = self._model()
+        await ts.put_state_dict(model.state_dict(), "fsdp")
+        return "ok"
+
+    @endpoint
+    async def pull_and_check(self):
+        model = self._model()  # same seed -> same reference weights
+        sd = model.state_dict()
+        with torch.no_grad():
+            for v in sd.values():
+                local = v.to_local() if hasattr(v, "to_local") else v
+                local.zero_()
+        out = await ts.get_state_dict("fsdp", sd)
+        ref = self._model().state_dict()
+        for k, v in out.items():
+            got = v.to_local() if hasattr(v, "to_local") else v
+            want = ref[k].to_local() if hasattr(ref[k], "to_local") else ref[k]
+            if not torch.equal(got, want):
+                return f"mismatch at {k}"
+        return "ok"
+
+
+@pytest.mark.parametrize("put_world,get_world", [(2, 2), (2, 3)])
+async def test_fully_shard_model_across_worlds(put_world, get_world):
+    """fully_shard (FSDP2) model pushed by one world, pulled — with
+    resharding — by an independent world of a different size."""
+    pytest.importorskip("transformers")
+    controller = await ts.initialize(
+        num_storage_volumes=put_world,
+        strategy=None,
+        storage_device="cpu",
+    )
+    put_mesh = get_mesh = None
+    try:
+        from torchstore_amd.strategy import LocalRankStrategy  # noqa: F401
+
+        pg1 = tempfile.mktemp(prefix=f"fsdp-pg-{uuid.uuid4().hex[:6]}")
+        put_mesh = await asyncio.to_thread(
+            spawn_actors, put_world, FsdpWorker, "fsdp-put",
+            put_world, pg1, controller, 123,
+        )
+        res = await put_mesh.push.call()
+        assert res == ["ok"] * put_world
+        pg2 = tempfile.mktemp(prefix=f"fsdp-pg-{uuid.uuid4().hex[:6]}")
+        get_mesh = await asyncio.to_thread(
+            spawn_actors, get_world, FsdpWorker, "fsdp-get",
+            get_world, pg2, controller, 123,
+        )
+        res = await get_mesh.pull_and_check.call()
+        assert res == ["ok"] * get_world, res
+    finally:
+        for m in (put_mesh, get_mesh):
+            if m is not None:
+                await m.stop()
+        await ts.shutdown()
+        await close_connections()
