@@ -99,9 +99,16 @@ def _typed_view(seg_u8: torch.Tensor, desc: ShmDescriptor) -> torch.Tensor:
     return seg_u8[: numel * esize].view(desc.dtype).reshape(desc.shape)
 
 
+def _pin_enabled() -> bool:
+    import os
+
+    # reference parity: TORCHSTORE_PIN_SHM opt-out (shared_memory.py:50-53)
+    return os.environ.get("TORCHSTORE_AMD_PIN_SHM", "1") != "0"
+
+
 def _try_pin(seg_u8: torch.Tensor, pinned: Dict[int, int]) -> None:
     """Pin a segment's pages for DMA; fail-open (copies still work unpinned)."""
-    if not torch.cuda.is_available():
+    if not torch.cuda.is_available() or not _pin_enabled():
         return
     ptr = seg_u8.untyped_storage().data_ptr()
     if ptr in pinned:
