@@ -497,20 +497,19 @@ class HipIpcTransportBuffer(TransportBuffer):
                 by_dev.setdefault(t.device.index, []).append((i, t))
             for dev, items in by_dev.items():
                 device = torch.device("cuda", dev)
+                sizes = [
+                    (t.numel() * t.element_size() + 255) & ~255
+                    for _i, t in items
+                ]
                 buf = None
                 off = 0
                 copies = []  # (src_view, dst_ptr) for one batched launch
-                for i, t in items:
-                    nb = t.numel() * t.element_size()
-                    aligned = (nb + 255) & ~255
+                for k, (i, t) in enumerate(items):
+                    aligned = sizes[k]
                     if buf is None or off + aligned > buf.numel():
                         if buf is not None:
                             self.pack_descs[-1] = (self.pack_descs[-1][0], off)
-                        remaining = sum(
-                            (tt.numel() * tt.element_size() + 255) & ~255
-                            for j, tt in items
-                            if (j, tt) == (i, t) or j > i
-                        )
+                        remaining = sum(sizes[k:])
                         size = min(self.PACK_BUF_CAP, max(remaining, aligned))
                         buf = torch.empty(size, dtype=torch.uint8, device=device)
                         self._packs.append(buf)
@@ -529,19 +528,22 @@ class HipIpcTransportBuffer(TransportBuffer):
                 rejects = gpu_ops.copy_views_to_ptrs(
                     copies, device, blocking=False
                 )
-                for t, _ptr in rejects:
-                    # kernel-inexpressible layout: pack via a contiguous copy
-                    tc = t.contiguous()
-                    idx = next(
-                        i for i, tt in pack_items if tt is t
-                    )
-                    pi, pk, poff = next(
-                        m for m in pack_meta if m[0] == idx
-                    )
+                if rejects:
                     from torchstore_amd.ops.slicing import byte_view
 
-                    nb = tc.numel() * tc.element_size()
-                    self._packs[pk][poff : poff + nb].copy_(byte_view(tc))
+                    # slot lookup by DESTINATION pointer: the same tensor
+                    # object may be packed under several keys (distinct
+                    # slots), so identity on the source would be ambiguous
+                    slot_by_ptr = {
+                        self._packs[pk].data_ptr() + poff: (pk, poff)
+                        for _pi, pk, poff in pack_meta
+                    }
+                    for t, ptr in rejects:
+                        # kernel-inexpressible layout: pack via torch copy
+                        tc = t.contiguous()
+                        pk, poff = slot_by_ptr[ptr]
+                        nb = tc.numel() * tc.element_size()
+                        self._packs[pk][poff : poff + nb].copy_(byte_view(tc))
 
         # every producing kernel AND every pack copy must be visible before
         # the volume's one-sided pulls read the staging memory from another
