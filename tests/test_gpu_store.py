@@ -470,12 +470,17 @@ async def test_packed_put_coalescing(monkeypatch):
     async def body():
         big = torch.randn(20 << 20, device="cuda")  # 80 MB > threshold
         strided = torch.randn(128, 64, device="cuda").t()  # stride(-1) != 1
+        shared = torch.randn(33, device="cuda")
         items = {
             "small0": torch.randn(300, 301, device="cuda"),
             "small1": torch.randn(7, device="cuda", dtype=torch.bfloat16),
             "zero_d": torch.tensor(3.5, device="cuda"),
             "big": big,
             "strided": strided,
+            # the SAME tensor object under two keys: both pack slots must
+            # be filled (slot lookup is by destination pointer)
+            "dup_a": shared,
+            "dup_b": shared,
             "obj": {"meta": 42},
         }
         await ts.put_batch(items)
