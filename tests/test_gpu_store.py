@@ -470,7 +470,9 @@ async def test_packed_put_coalescing(monkeypatch):
     async def body():
         big = torch.randn(20 << 20, device="cuda")  # 80 MB > threshold
         strided = torch.randn(128, 64, device="cuda").t()  # stride(-1) != 1
-        shared = torch.randn(33, device="cuda")
+        # strided AND shared: the kernel rejects the layout, and the
+        # torch-copy fallback must fill BOTH slots (matched by slot ptr)
+        shared = torch.randn(64, 33, device="cuda").t()
         items = {
             "small0": torch.randn(300, 301, device="cuda"),
             "small1": torch.randn(7, device="cuda", dtype=torch.bfloat16),
