@@ -1,6 +1,7 @@
 """Actor runtime: spawn, RPC, meshes, serialization, teardown."""
 
 import asyncio
+import os
 
 import pytest
 import torch
@@ -134,6 +135,28 @@ async def test_mesh_slice_and_concurrent_calls():
             *(mesh.handles[i % 4].add.call_one(i, i) for i in range(32))
         )
         assert outs == [2 * i for i in range(32)]
+    finally:
+        await mesh.stop()
+        await close_connections()
+
+
+@pytest.mark.skipif(
+    os.environ.get("TORCHSTORE_AMD_SLOW_TESTS", "0") != "1",
+    reason="multi-GB frame; TORCHSTORE_AMD_SLOW_TESTS=1 enables",
+)
+async def test_giant_rpc_frame_over_2gib():
+    """The RPC framing is 64-bit end to end: a single >2^31-byte tensor
+    payload crosses the wire intact (the reference needs a frame-size env
+    override for this — HYPERACTOR_CODEC_MAX_FRAME_LENGTH)."""
+    t = torch.arange(1 << 29, dtype=torch.int64)  # 4 GiB
+    mesh = spawn_actors(1, Echo, "giant")
+    try:
+        h = mesh.handles[0]
+        assert tuple(await h.put.call_one("g", t)) == (1 << 29,)
+        out = await h.get.call_one("g")
+        assert out.shape == t.shape and out.dtype == t.dtype
+        assert torch.equal(out[:1000], t[:1000])
+        assert torch.equal(out[-1000:], t[-1000:])
     finally:
         await mesh.stop()
         await close_connections()

@@ -188,3 +188,35 @@ def test_spmd_multihost_volume_placement(strategy_name):
             assert result["volume_ids"] == ["fakehost0", "fakehost1"]
         else:
             assert result["volume_ids"] == ["0", "1"]
+
+
+@pytest.mark.skipif(
+    os.environ.get("TORCHSTORE_AMD_SLOW_TESTS", "0") != "1",
+    reason="4-process bring-up is slow; TORCHSTORE_AMD_SLOW_TESTS=1 enables",
+)
+def test_spmd_two_hosts_two_ranks_each():
+    """2 fake hosts x 2 ranks: per-host volume spawns with global id
+    offsets (rank_offset seed) + cross-host gets."""
+    world, local_world = 4, 2
+    port = pick_free_port()
+    ctx = mp.get_context("spawn")
+    procs, paths = [], []
+    for rank in range(world):
+        path = tempfile.mktemp(prefix=f"spmd-mh4-{rank}")
+        paths.append(path)
+        p = ctx.Process(
+            target=_multihost_worker,
+            args=(rank, world, local_world, port, path, "local_rank"),
+        )
+        p.start()
+        procs.append(p)
+    for p in procs:
+        p.join(timeout=300)
+        assert not p.is_alive(), "worker hung"
+    for rank, path in enumerate(paths):
+        with open(path) as f:
+            result = json.load(f)
+        assert "error" not in result, f"rank {rank}: {result.get('error')}"
+        assert result["ok"], f"rank {rank} read wrong peer data"
+        assert result["volume_ids"] == ["0", "1", "2", "3"], result
+        assert result["hostnames"] == ["fakehost0", "fakehost1"], result
