@@ -205,3 +205,41 @@ async def test_repush_with_new_mesh_replaces_old_epoch():
                 await m.stop()
         await ts.shutdown()
         await close_connections()
+
+
+async def test_fully_replicated_dtensor_demoted_to_plain():
+    """Reference EP/MoE semantics (test_tensor_slice.py:400-506): a fully
+    Replicate DTensor is stored as a PLAIN tensor — no commit gate, no
+    shard bookkeeping — and reads back as the full tensor immediately."""
+    controller = await ts.initialize(
+        num_storage_volumes=2,
+        strategy=LocalRankStrategy(),
+        storage_device="cpu",
+    )
+    put_mesh = None
+    try:
+        put_mesh = await _spawn_world(2, controller, "rep")
+        # ONLY rank 0 puts (ranks_to_skip semantics): a replicated entry
+        # must be readable without every coordinate committing
+        results = await asyncio.gather(
+            put_mesh.handles[0].put_dtensor.call_one(
+                "e", (8, 8), (2,), ["r"], False
+            ),
+            put_mesh.handles[1].put_dtensor.call_one(
+                "e", (8, 8), (2,), ["r"], True
+            ),
+        )
+        assert results == ["ok", "skipped"]
+        out = await ts.get("e")  # readable despite rank 1 skipping
+        assert out.shape == (8, 8)
+        # index kind: plain TENSOR, not TENSOR_SLICE
+        from torchstore_amd.controller import ObjectType
+
+        located = await controller.locate.call_one(["e"])
+        kinds = {i.object_type for i in located["e"].values()}
+        assert kinds == {ObjectType.TENSOR}
+    finally:
+        if put_mesh is not None:
+            await put_mesh.stop()
+        await ts.shutdown()
+        await close_connections()

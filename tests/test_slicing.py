@@ -126,3 +126,35 @@ def test_expected_coordinates():
     coords = set(TensorSlice.expected_coordinates((2, 3)))
     assert len(coords) == 6
     assert (1, 2) in coords
+
+
+def test_union_volume_and_disjoint():
+    import torch
+
+    from torchstore_amd.ops.slicing import (
+        assemble,
+        regions_disjoint,
+        union_volume,
+    )
+
+    a = ((0, 0), (4, 4))
+    b = ((2, 0), (4, 4))   # overlaps a by 2x4
+    c = ((6, 0), (2, 4))
+    assert not regions_disjoint([a, b])
+    assert regions_disjoint([a, c])
+    assert union_volume([a, b]) == 4 * 4 + 4 * 4 - 2 * 4
+    assert union_volume([a, b, c]) == 24 + 8
+    # overlapping parts that leave a GAP inside the bounding box must be
+    # rejected even when raw numel sums look sufficient (round-1's sum
+    # check could be fooled by overlap compensating a gap)
+    import pytest as _pytest
+
+    p1 = ((0,), torch.ones(4))     # [0,4)
+    p2 = ((2,), torch.ones(4))     # [2,6) — overlaps p1
+    p3 = ((7,), torch.ones(1))     # [7,8) — leaves gap [6,7)
+    # box [0,8)=8 elements; numels sum to 9 but union covers only 7
+    with _pytest.raises(ValueError, match="do not tile"):
+        assemble([p1, p2, p3])
+    # the same parts WITHOUT the gap assemble fine
+    out, origin = assemble([p1, p2, ((6,), torch.ones(2))])
+    assert origin == (0,) and out.numel() == 8
