@@ -1,0 +1,88 @@
+"""Request / TensorSlice wire types (reference test_tensor_slice.py):
+DTensor → Request building, trivially-local demotion (the EP/MoE case),
+meta stripping, LocalShard passthrough."""
+
+import tempfile
+
+import pytest
+import torch
+
+from torchstore_amd.types import LocalShard, Request, TensorSlice
+
+
+def _pg1():
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        dist.init_process_group(
+            "gloo", init_method=f"file://{tempfile.mktemp()}",
+            rank=0, world_size=1,
+        )
+    return dist
+
+
+def test_mesh1_and_replicate_dtensors_demote_to_plain():
+    """Reference types.py:58-85,145-151: a DTensor whose local tensor IS
+    the full tensor (mesh size 1, or all-Replicate) stores as a PLAIN
+    tensor — no TensorSlice, no commit gate."""
+    dist = _pg1()
+    try:
+        from torch.distributed.device_mesh import init_device_mesh
+        from torch.distributed.tensor import (
+            Replicate,
+            Shard,
+            distribute_tensor,
+        )
+
+        mesh = init_device_mesh("cpu", (1,))
+        t = torch.randn(8, 4)
+        for placements in ([Shard(0)], [Shard(1)], [Replicate()]):
+            dt = distribute_tensor(t.clone(), mesh, placements)
+            req = Request.from_any("k", dt)
+            assert req.tensor_slice is None, placements
+            assert torch.equal(req.tensor_val, t)
+            assert not req.is_object
+    finally:
+        dist.destroy_process_group()
+
+
+def test_localshard_and_plain_tensor_requests():
+    t = torch.randn(6, 6)
+    req = Request.from_any("k", t)
+    assert req.tensor_slice is None and req.tensor_val is t
+
+    s = TensorSlice((0, 0), (3, 6), (6, 6), (0,), (2,))
+    req2 = Request.from_any("k", LocalShard(tensor=t[:3], slice=s))
+    assert req2.tensor_slice == s
+    assert torch.equal(req2.tensor_val, t[:3])
+
+    req3 = Request.from_any("k", {"cfg": 1})
+    assert req3.is_object and req3.objects == {"cfg": 1}
+
+
+def test_meta_only_strips_payload():
+    t = torch.randn(4)
+    req = Request.from_any("k", t)
+    m = req.meta_only()
+    assert m.tensor_val is None and m.key == "k"
+    assert not m.has_payload
+    assert req.tensor_val is t  # original untouched
+
+    obj = Request.from_any("k", [1, 2, 3])
+    mo = obj.meta_only()
+    assert mo.is_object and mo.objects is None
+
+
+def test_tensor_slice_normalization_and_hash():
+    a = TensorSlice([0, 2], [4, 4], [8, 8], [1], [2])
+    b = TensorSlice((0, 2), (4, 4), (8, 8), (1,), (2,))
+    assert a == b and hash(a) == hash(b)
+    assert a.offsets == (0, 2) and isinstance(a.offsets[0], int)
+    assert a.numel() == 16
+
+
+def test_expected_coordinates_cartesian():
+    coords = list(TensorSlice.expected_coordinates((2, 3)))
+    assert len(coords) == 6
+    assert tuple(coords[0]) == (0, 0) and tuple(coords[-1]) == (1, 2)
+    assert list(TensorSlice.expected_coordinates(())) == [()]
