@@ -347,3 +347,25 @@ async def test_reput_across_volumes_serves_newest(monkeypatch):
         assert (await ts.get("h")).eq(7.0).all()
     finally:
         await ts.shutdown()
+
+
+async def test_noncontiguous_put(transport):
+    """Non-contiguous values (transposes, slices) store their materialized
+    content (reference test_store.py:556)."""
+
+    async def body():
+        base = torch.randn(32, 48)
+        views = {
+            "nc/t": base.t(),              # stride-swapped
+            "nc/slice": base[::2, 1:17],   # offset + both dims strided
+            "nc/chan": base.unsqueeze(0).expand(3, 32, 48)[1],
+        }
+        await ts.put_batch(views)
+        for k, v in views.items():
+            out = await ts.get(k)
+            assert torch.equal(out, v.contiguous()), k
+            dest = torch.zeros(v.shape)
+            await ts.get(k, dest)
+            assert torch.equal(dest, v), k
+
+    await _with_store(transport, body)
