@@ -264,3 +264,80 @@ def test_pg_pair_retry_builds_fresh_rendezvous(monkeypatch):
     fresh = cache.get_or_create("v1", "gloo")
     assert fresh.info.pair_id != first_id
     assert fresh.info.port != 0
+
+
+@pytest.fixture
+def pushpull_env(chunked_env, monkeypatch):
+    """chunked_env plus a faked try_export (push/pull use it volume-side)."""
+    monkeypatch.setattr(
+        hip_ipc, "try_export", lambda t, generation=None: _fake_export(t)
+    )
+    return chunked_env
+
+
+def test_push_put_roundtrip(pushpull_env):
+    """Direct push: volume exports payloads, client writes them in place,
+    volume_receive adopts; pending state fully consumed."""
+    buffer, volume, volume_ctx = pushpull_env
+    cache: ChunkStagingCache = volume_ctx.cache(ChunkStagingCache)
+    t1 = torch.arange(100, dtype=torch.uint8)
+    t2 = torch.arange(50, dtype=torch.uint8) + 7
+    reqs = [Request(key="a"), Request(key="b")]
+    payload = [("pending", None), ("pending", None)]
+    asyncio.run(buffer._push_put([(0, t1), (1, t2)], reqs, payload))
+    assert payload[0][0] == "pushed" and payload[1][0] == "pushed"
+    assert len(cache.push_pending) == 1
+    buffer.payload = payload
+    out = asyncio.run(
+        buffer.volume_receive(reqs, [None, None], torch.device("cpu"))
+    )
+    assert torch.equal(out[0], t1) and torch.equal(out[1], t2)
+    assert not cache.push_pending, "consumed tokens must clear"
+
+
+def test_push_abort_releases_pending(pushpull_env, monkeypatch):
+    buffer, volume, volume_ctx = pushpull_env
+    cache: ChunkStagingCache = volume_ctx.cache(ChunkStagingCache)
+
+    def boom(copies):
+        raise RuntimeError("injected copy failure")
+
+    monkeypatch.setattr(hip_ipc, "_run_copies", boom)
+    t = torch.ones(64, dtype=torch.uint8)
+    with pytest.raises(RuntimeError, match="injected"):
+        asyncio.run(
+            buffer._push_put([(0, t)], [Request(key="a")], [("pending", None)])
+        )
+    assert not cache.push_pending, "aborted push leaked pending payloads"
+
+
+def test_pull_get_roundtrip_and_stash_release(pushpull_env):
+    buffer, volume, volume_ctx = pushpull_env
+    cache: ChunkStagingCache = volume_ctx.cache(ChunkStagingCache)
+    stored = torch.arange(200, dtype=torch.uint8)
+    volume.store = _FakeStore(stored)
+    dest = torch.zeros_like(stored)
+    payload = {0: None}
+    asyncio.run(
+        buffer._batched_pull([(0, Request(key="k"), dest)], payload)
+    )
+    assert payload[0] == ("pulled", None)
+    assert torch.equal(dest, stored)
+    assert not cache.pull_stash, "pull_release must clear the stash"
+
+
+def test_pull_abort_releases_stash(pushpull_env, monkeypatch):
+    buffer, volume, volume_ctx = pushpull_env
+    cache: ChunkStagingCache = volume_ctx.cache(ChunkStagingCache)
+    volume.store = _FakeStore(torch.ones(64, dtype=torch.uint8))
+
+    def boom(copies):
+        raise RuntimeError("injected pull copy failure")
+
+    monkeypatch.setattr(hip_ipc, "_run_copies", boom)
+    dest = torch.zeros(64, dtype=torch.uint8)
+    with pytest.raises(RuntimeError, match="injected"):
+        asyncio.run(
+            buffer._batched_pull([(0, Request(key="k"), dest)], {0: None})
+        )
+    assert not cache.pull_stash

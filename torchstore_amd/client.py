@@ -63,6 +63,70 @@ class _CachedFetchPlan:
     busy: bool = False  # a concurrent identical call must not share subs
 
 
+def _split_huge_tensors(
+    requests: List[Request], ref: StorageVolumeRef
+) -> List[Request]:
+    """Split GPU-resident plain tensors whose allocator block is too large
+    to IPC-export (>=2 GiB, the platform dmabuf-import limit) into
+    row-range shards with synthetic TensorSlices.
+
+    The pieces are ordinary shards: they store, commit (all coordinates
+    land in one notify batch), reshard and reassemble through the existing
+    machinery — but each piece moves over the DIRECT one-sided push/pull
+    paths instead of the double-staged windowed protocol (measured ~870
+    GB/s windowed vs ~2.3 TB/s direct at these sizes).  Only plain
+    tensors split; user DTensor shards keep their real coordinates and
+    use the windowed fallback when oversized.
+    """
+    from torchstore_amd.transport.hip_ipc import IPC_BLOCK_LIMIT
+
+    if not str(ref.device).startswith("cuda"):
+        return requests
+    out: List[Request] = []
+    for r in requests:
+        t = r.tensor_val
+        if (
+            r.is_object
+            or t is None
+            or r.tensor_slice is not None
+            or t.device.type != "cuda"
+            or t.dim() == 0
+            or t.numel() * t.element_size() < IPC_BLOCK_LIMIT
+        ):
+            out.append(r)
+            continue
+        tc = t.contiguous()
+        rows = tc.shape[0]
+        row_bytes = tc.numel() * tc.element_size() // max(rows, 1)
+        if rows < 2 or row_bytes == 0 or row_bytes >= IPC_BLOCK_LIMIT:
+            out.append(r)  # single-row giants keep the windowed path
+            continue
+        rows_per_piece = max(1, (1 << 30) // row_bytes)
+        k = (rows + rows_per_piece - 1) // rows_per_piece
+        if k < 2:
+            out.append(r)
+            continue
+        gshape = tuple(tc.shape)
+        for j in range(k):
+            r0 = j * rows_per_piece
+            nr = min(rows_per_piece, rows - r0)
+            out.append(
+                Request(
+                    key=r.key,
+                    tensor_val=tc[r0 : r0 + nr],
+                    tensor_slice=TensorSlice(
+                        offsets=(r0,) + (0,) * (tc.dim() - 1),
+                        local_shape=(nr,) + gshape[1:],
+                        global_shape=gshape,
+                        coordinates=(j,),
+                        mesh_shape=(k,),
+                    ),
+                    inplace=r.inplace,
+                )
+            )
+    return out
+
+
 def _full_region_slice(global_shape: Sequence[int]) -> TensorSlice:
     shape = tuple(global_shape)
     return TensorSlice(
@@ -117,6 +181,7 @@ class LocalClient:
         requests = [Request.from_any(k, v) for k, v in items.items()]
         volume_id = self._strategy.select_volume_id(list(volumes.keys()))
         ref = self._volume_ref(volume_id)
+        requests = _split_huge_tensors(requests, ref)
         tracker = LatencyTracker(f"put_batch[{len(requests)}]")
         from torchstore_amd.utils.logging import roctx_range
 

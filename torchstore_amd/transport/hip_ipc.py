@@ -21,12 +21,21 @@ Mechanics (native side in ``csrc/hipstore.hip``):
 
 **≥2 GiB blocks.** ``hipIpcOpenMemHandle`` of a ≥2³¹-byte dmabuf hangs on
 this platform (measured — the export succeeds, the peer's import never
-returns).  Tensors living in such blocks take a *windowed* path instead:
-the volume owns a pool of <2 GiB staging chunks (exported once); the client
-opens the staging and copies windows in/out of it over xGMI, with a
-handshake RPC per window committing each one.  Staging chunks assigned to
-an operation are returned to the pool when the operation's final RPC lands
-(a client that dies mid-transfer leaks its chunk until ``reset``).
+returns).  Three escalating strategies handle tensors touching such
+blocks:
+
+1. huge PLAIN tensors are auto-split client-side into ~1 GiB row shards
+   (``client._split_huge_tensors``) so every piece is exportable;
+2. a sub-2 GiB tensor merely LIVING in an oversized block (a view of a
+   huge source or destination) moves DIRECTLY with the roles flipped:
+   puts PUSH into volume-exported payloads, gets PULL from token-pinned
+   volume exports — one handshake RPC pair per batch, no staging hop;
+3. only when the volume's own side would be ≥2 GiB (an unsplit sharded
+   local) does the *windowed* fallback run: a pool of <2 GiB staging
+   chunks, 3-deep pipelined (client copies ∥ volume copies ∥ commit
+   RPCs).  Staging chunks return to the pool when the operation's final
+   RPC lands (a client that dies mid-transfer leaks its chunk until
+   ``reset``).
 
 CPU tensors and objects in a batch ride inline in the RPC frame (an IPC
 batch can be mixed — e.g. a state_dict with scalar stats).
@@ -161,6 +170,11 @@ class ChunkStagingCache(TransportCache):
         ] = {}
         # async put-commit state: token -> (torch stream, last event)
         self.op_streams: Dict[str, Tuple[Any, Any]] = {}
+        # direct push/pull state (no staging hop):
+        #   push: token -> payload tensors the CLIENT writes into directly
+        #   pull: token -> value tensors pinned alive while the client reads
+        self.push_pending: Dict[str, List[Optional[torch.Tensor]]] = {}
+        self.pull_stash: Dict[str, List[Optional[torch.Tensor]]] = {}
 
     def acquire(
         self,
@@ -241,6 +255,8 @@ class ChunkStagingCache(TransportCache):
             self.finish(token)
         self.free.clear()
         self.by_token.clear()
+        self.push_pending.clear()
+        self.pull_stash.clear()
 
 
 def _run_copies(copies: List[Tuple[int, int, int, int, int]]) -> None:
@@ -449,6 +465,73 @@ class HipIpcTransportBuffer(TransportBuffer):
         if phase == "chunk_release":
             cache.release(args)
             return "ok"
+        if phase == "push_alloc":
+            # direct put of client tensors living in unexportable (>=2 GiB)
+            # blocks: the VOLUME exports the (<2 GiB) payload tensors and
+            # the client writes into them one-sided — no staging hop, no
+            # per-window RPCs (the windowed path remains the fallback for
+            # payloads that are themselves >=2 GiB)
+            token, items = args
+            store = getattr(volume, "store", None)
+            payloads: List[Optional[torch.Tensor]] = []
+            descs: List[Optional[IpcDescriptor]] = []
+            for meta, shape, dtype in items:
+                payload = None
+                if meta is not None and store is not None:
+                    prior = store.find_existing(meta)
+                    if (
+                        prior is not None
+                        and tuple(prior.shape) == tuple(shape)
+                        and prior.dtype == dtype
+                        and prior.is_contiguous()
+                        and prior.device == device
+                    ):
+                        payload = prior
+                if payload is None:
+                    payload = torch.empty(shape, dtype=dtype, device=device)
+                desc = try_export(payload)
+                if desc is None:
+                    payloads.append(None)  # unexportable payload: fallback
+                    descs.append(None)
+                else:
+                    payloads.append(payload)
+                    descs.append(desc)
+            cache.push_pending[token] = payloads
+            return descs
+        if phase == "push_release":
+            cache.push_pending.pop(args, None)
+            return "ok"
+        if phase == "pull_init":
+            # direct get into client destinations living in unexportable
+            # blocks: the volume exports the stored values (packing strided
+            # ones) and PINS them under the token while the client reads
+            token, metas = args
+            from torchstore_amd.ops import gpu as gpu_ops
+
+            stash: List[Optional[torch.Tensor]] = []
+            descs = []
+            launched = False
+            for meta in metas:
+                value = volume.store.fetch(meta)
+                if not isinstance(value, torch.Tensor) or value.device != device:
+                    stash.append(None)
+                    descs.append(None)
+                    continue
+                if value.is_contiguous():
+                    vc = value
+                else:
+                    vc = gpu_ops.pack_region(value)
+                    launched = True
+                desc = try_export(vc)
+                stash.append(vc if desc is not None else None)
+                descs.append(desc)
+            if launched:
+                torch.cuda.current_stream(device).synchronize()
+            cache.pull_stash[token] = stash
+            return descs
+        if phase == "pull_release":
+            cache.pull_stash.pop(args, None)
+            return "ok"
         raise ValueError(f"unknown handshake phase {phase!r}")
 
     # ------------------------------------------------------------- put --
@@ -568,14 +651,65 @@ class HipIpcTransportBuffer(TransportBuffer):
                 torch.cuda.current_stream(tc.device).synchronize()
                 staged.append((pi, tc))
 
+        push_items: List[Tuple[int, torch.Tensor]] = []
         for i, tc in staged:
             desc = try_export(tc, gens.get(tc.device.index))
-            if desc is None:
+            if desc is not None:
+                payload[i] = ("ipc", desc)
+            elif tc.numel() * tc.element_size() < IPC_BLOCK_LIMIT:
+                # the tensor itself fits an exportable block but LIVES in
+                # an unexportable (>=2 GiB) one (e.g. a view of a huge
+                # source): PUSH — the volume exports its payload and the
+                # client writes into it directly, no staging hop
+                push_items.append((i, tc))
+            else:
                 token = await self._chunked_put_windows(tc, requests[i])
                 payload[i] = ("chunked", token)
-            else:
-                payload[i] = ("ipc", desc)
+        if push_items:
+            await self._push_put(push_items, requests, payload)
         self.payload = payload
+
+    async def _push_put(
+        self,
+        push_items: List[Tuple[int, torch.Tensor]],
+        requests: Sequence[Request],
+        payload: List[Tuple[str, Any]],
+    ) -> None:
+        volume = self._volume_ref.volume
+        cache: IpcOpenCache = self._client_ctx.cache(IpcOpenCache)
+        token = uuid.uuid4().hex
+        items = [
+            (requests[i].meta_only(), tuple(tc.shape), tc.dtype)
+            for i, tc in push_items
+        ]
+        descs = await volume.handshake.call_one(
+            self, (token, items), "push_alloc"
+        )
+        try:
+            copies = []
+            chunk_fallback: List[int] = []
+            for j, ((i, tc), desc) in enumerate(zip(push_items, descs)):
+                if desc is None:
+                    chunk_fallback.append(j)
+                    continue
+                dst_ptr = cache.resolve(desc, tc.device.index)
+                copies.append(
+                    (dst_ptr, desc.device_index, tc.data_ptr(),
+                     tc.device.index, tc.numel() * tc.element_size())
+                )
+                payload[i] = ("pushed", (token, j))
+            if copies:
+                await asyncio.to_thread(_run_copies, copies)
+            for j in chunk_fallback:
+                i, tc = push_items[j]
+                tok = await self._chunked_put_windows(tc, requests[i])
+                payload[i] = ("chunked", tok)
+        except BaseException:
+            try:
+                await volume.handshake.call_one(self, token, "push_release")
+            except Exception:  # noqa: BLE001
+                pass
+            raise
 
     async def volume_receive(self, requests, existing, device):
         cache: IpcOpenCache = self._volume_ctx.cache(IpcOpenCache)
@@ -598,6 +732,16 @@ class HipIpcTransportBuffer(TransportBuffer):
                 if payload is None:
                     raise RuntimeError("chunked put token unknown")
                 out[i] = payload
+                continue
+            if kind == "pushed":
+                token, j = value
+                pending = chunks.push_pending.get(token)
+                if pending is None or pending[j] is None:
+                    raise RuntimeError("pushed put token unknown")
+                out[i] = pending[j]
+                pending[j] = None
+                if all(p is None for p in pending):
+                    chunks.push_pending.pop(token, None)
                 continue
             if kind == "packed":
                 pk, off, shape, dtype = value
@@ -674,7 +818,12 @@ class HipIpcTransportBuffer(TransportBuffer):
         return -1
 
     async def _stage_get_normal(
-        self, i: int, r: Request, synced: set, gens: Optional[Dict] = None
+        self,
+        i: int,
+        r: Request,
+        synced: set,
+        gens: Optional[Dict] = None,
+        pull_collect: Optional[List] = None,
     ) -> Tuple[str, Any]:
         """Per-request direct staging: export the dest (or a dense scratch),
         falling back to the windowed path for >=2 GiB blocks."""
@@ -702,15 +851,73 @@ class HipIpcTransportBuffer(TransportBuffer):
                 gen = gpu_ops.alloc_generation(di)
                 gens[di] = gen
         desc = try_export(target, gen)
-        if desc is None:
-            token = await self._chunked_get_windows(r, target)
-            return ("chunked", token)
-        return ("ipc", desc)
+        if desc is not None:
+            return ("ipc", desc)
+        # dest lives in an unexportable (>=2 GiB) block: PULL — the volume
+        # exports the stored value (pinned under a token) and the client
+        # copies it locally; windowed staging only if the volume can't
+        # export its side either.  Pulls for one operation batch under a
+        # single token (one RPC pair, one copy batch).
+        if pull_collect is not None:
+            pull_collect.append((i, r, target))
+            return ("pending_pull", None)
+        # no collector (direct call): run a one-entry batch immediately
+        single: Dict[int, Tuple[str, Any]] = {}
+        await self._batched_pull([(i, r, target)], single)
+        return single[i]
+
+    async def _batched_pull(
+        self,
+        entries: List[Tuple[int, Request, torch.Tensor]],
+        payload,
+    ) -> None:
+        """One pull_init / pull_release RPC pair for all entries; entries
+        the volume cannot export fall back to the windowed path."""
+        volume = self._volume_ref.volume
+        cache: IpcOpenCache = self._client_ctx.cache(IpcOpenCache)
+        token = uuid.uuid4().hex
+        metas = [r.meta_only() for _i, r, _t in entries]
+        descs = await volume.handshake.call_one(
+            self, (token, metas), "pull_init"
+        )
+        fallback: List[Tuple[int, Request, torch.Tensor]] = []
+        try:
+            copies = []
+            for (i, r, target), desc in zip(entries, descs):
+                if desc is None:
+                    fallback.append((i, r, target))
+                    continue
+                if (
+                    desc.dtype != target.dtype
+                    or desc.nbytes != target.numel() * target.element_size()
+                ):
+                    raise RuntimeError(
+                        f"pull mismatch for {r.key}: stored {desc.shape} "
+                        f"{desc.dtype} vs dest {tuple(target.shape)} "
+                        f"{target.dtype}"
+                    )
+                src_ptr = cache.resolve(desc, target.device.index)
+                copies.append(
+                    (target.data_ptr(), target.device.index, src_ptr,
+                     desc.device_index, desc.nbytes)
+                )
+                payload[i] = ("pulled", None)
+            if copies:
+                await asyncio.to_thread(_run_copies, copies)
+        finally:
+            try:
+                await volume.handshake.call_one(self, token, "pull_release")
+            except Exception:  # noqa: BLE001
+                pass
+        for i, r, target in fallback:
+            tok = await self._chunked_get_windows(r, target)
+            payload[i] = ("chunked", tok)
 
     async def client_stage_get(self, requests: Sequence[Request]) -> None:
         payload: List[Tuple[str, Any]] = []
         synced: set = set()
         gens: Dict[int, int] = {}  # per-device allocator generation, per batch
+        pulls: List[Tuple[int, Request, torch.Tensor]] = []
         vol_dev = self._volume_device_index()
         # cross-device small/strided pieces coalesce through ONE bounce
         # buffer per op: the volume packs them locally (one kernel), moves
@@ -734,7 +941,9 @@ class HipIpcTransportBuffer(TransportBuffer):
                 payload.append(("bounce", None))  # offsets assigned below
                 bounce_plan.append((i, nbytes))
                 continue
-            payload.append(await self._stage_get_normal(i, r, synced, gens))
+            payload.append(
+                await self._stage_get_normal(i, r, synced, gens, pulls)
+            )
 
         self.bounce_descs: List[IpcDescriptor] = []
         self._bounces: List[torch.Tensor] = []
@@ -747,7 +956,9 @@ class HipIpcTransportBuffer(TransportBuffer):
             oversized = [e for e in aligned if e[2] > cap]
             aligned = [e for e in aligned if e[2] <= cap]
             for i, _n, _a in oversized:
-                payload[i] = await self._stage_get_normal(i, requests[i], synced, gens)
+                payload[i] = await self._stage_get_normal(
+                    i, requests[i], synced, gens, pulls
+                )
             remaining = sum(a for _, _, a in aligned)
             off = 0
             size = 0
@@ -776,10 +987,12 @@ class HipIpcTransportBuffer(TransportBuffer):
                 self.bounce_descs = []
                 for i, nbytes, a in aligned:
                     payload[i] = await self._stage_get_normal(
-                        i, requests[i], synced, gens
+                        i, requests[i], synced, gens, pulls
                     )
             elif device.index not in synced:
                 torch.cuda.current_stream(device).synchronize()
+        if pulls:
+            await self._batched_pull(pulls, payload)
         self.payload = payload
 
     async def volume_send(self, requests, values):
@@ -804,6 +1017,10 @@ class HipIpcTransportBuffer(TransportBuffer):
             if kind == "chunked":
                 # windows already delivered during the handshake phases
                 chunks.release(value)
+                reply.append(("done", None))
+                continue
+            if kind == "pulled":
+                # the client already copied out of the pinned export
                 reply.append(("done", None))
                 continue
             if kind == "bounce":
