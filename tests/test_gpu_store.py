@@ -700,3 +700,57 @@ async def test_tiered_store_hbm_spill_to_host():
         assert torch.equal(got, c.cpu())
     finally:
         await ts.shutdown()
+
+
+@requires_gpu
+async def test_huge_tensor_autosplit_region_fetch():
+    """A >=2 GiB plain tensor auto-splits into exportable row shards;
+    whole-tensor and piece-boundary-crossing region fetches both work."""
+    from torchstore_amd.controller import ObjectType
+    from torchstore_amd.types import LocalShard, TensorSlice
+
+    async def body():
+        rows = 300_000_000  # 2.4 GB f64... f32 x2 cols
+        t = torch.empty(rows, 2, device="cuda")
+        t[:, 0] = torch.arange(rows, device="cuda", dtype=torch.float32)
+        t[:, 1] = 1.0
+        await ts.put("huge", t)
+        c = ts.client()
+        located = await c._controller.locate.call_one(["huge"])
+        infos = list(located["huge"].values())
+        slices = set()
+        for i in infos:
+            assert i.object_type == ObjectType.TENSOR_SLICE
+            slices |= i.tensor_slices
+        assert len(slices) >= 2, "expected an auto-split shard layout"
+        boundary = min(
+            s.offsets[0] for s in slices if s.offsets[0] > 0
+        )
+        # region crossing the split boundary assembles from two pieces
+        lo = boundary - 50
+        dest = LocalShard(
+            tensor=torch.zeros(100, 2, device="cuda"),
+            slice=TensorSlice(
+                offsets=(lo, 0), local_shape=(100, 2),
+                global_shape=(rows, 2), coordinates=(), mesh_shape=(),
+            ),
+        )
+        await ts.get("huge", dest)
+        torch.cuda.synchronize()
+        expect = torch.arange(lo, lo + 100, device="cuda",
+                              dtype=torch.float32)
+        assert torch.equal(dest.tensor[:, 0], expect)
+        assert dest.tensor[:, 1].eq(1.0).all()
+        # full-tensor in-place get (dest in an unexportable block -> PULL)
+        full = torch.zeros_like(t)
+        await ts.get("huge", full)
+        torch.cuda.synchronize()
+        assert torch.equal(full[::1_000_000], t[::1_000_000])
+        # in-place re-put (push_alloc reuses stored pieces) + re-read
+        t[:, 1] = 2.0
+        await ts.put("huge", t)
+        await ts.get("huge", full)
+        torch.cuda.synchronize()
+        assert full[::997][:, 1].eq(2.0).all()
+
+    await _with_store(body, transport=TransportType.HIP_IPC)
