@@ -737,10 +737,9 @@ async def test_huge_tensor_autosplit_region_fetch():
         )
         await ts.get("huge", dest)
         torch.cuda.synchronize()
-        expect = torch.arange(lo, lo + 100, device="cuda",
-                              dtype=torch.float32)
-        assert torch.equal(dest.tensor[:, 0], expect)
-        assert dest.tensor[:, 1].eq(1.0).all()
+        # compare against the SOURCE rows (f32 arange is inexact above
+        # 2^24, so an independently computed arange would diverge)
+        assert torch.equal(dest.tensor, t[lo : lo + 100])
         # full-tensor in-place get (dest in an unexportable block -> PULL)
         full = torch.zeros_like(t)
         await ts.get("huge", full)
