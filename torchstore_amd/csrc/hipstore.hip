@@ -362,6 +362,42 @@ copy_slices_kernel(const SliceDesc* __restrict__ descs, uint32_t nslices,
       const uint32_t rb4 = d.row_bytes >> 4;
       const uint32_t elems = nrows * rb4;
       uint32_t i = lane;
+      if ((rb4 & (rb4 - 1)) == 0) {
+        // power-of-two row: shift/mask replaces the per-element udiv pair
+        // (~30 cycles each — visible in the LLC-resident regime)
+        const uint32_t sh = 31 - __clz(rb4);
+        const uint32_t msk = rb4 - 1;
+        for (; i + 192 < elems; i += 256) {
+          uint32_t i0 = i, i1 = i + 64, i2 = i + 128, i3 = i + 192;
+          const uint4* s0 = reinterpret_cast<const uint4*>(
+              sbase + (int64_t)(i0 >> sh) * sstride + ((i0 & msk) << 4));
+          const uint4* s1 = reinterpret_cast<const uint4*>(
+              sbase + (int64_t)(i1 >> sh) * sstride + ((i1 & msk) << 4));
+          const uint4* s2 = reinterpret_cast<const uint4*>(
+              sbase + (int64_t)(i2 >> sh) * sstride + ((i2 & msk) << 4));
+          const uint4* s3 = reinterpret_cast<const uint4*>(
+              sbase + (int64_t)(i3 >> sh) * sstride + ((i3 & msk) << 4));
+          uint4 v0 = *s0;
+          uint4 v1 = *s1;
+          uint4 v2 = *s2;
+          uint4 v3 = *s3;
+          *reinterpret_cast<uint4*>(
+              dbase + (int64_t)(i0 >> sh) * dstride + ((i0 & msk) << 4)) = v0;
+          *reinterpret_cast<uint4*>(
+              dbase + (int64_t)(i1 >> sh) * dstride + ((i1 & msk) << 4)) = v1;
+          *reinterpret_cast<uint4*>(
+              dbase + (int64_t)(i2 >> sh) * dstride + ((i2 & msk) << 4)) = v2;
+          *reinterpret_cast<uint4*>(
+              dbase + (int64_t)(i3 >> sh) * dstride + ((i3 & msk) << 4)) = v3;
+        }
+        for (; i < elems; i += 64) {
+          *reinterpret_cast<uint4*>(
+              dbase + (int64_t)(i >> sh) * dstride + ((i & msk) << 4)) =
+              *reinterpret_cast<const uint4*>(
+                  sbase + (int64_t)(i >> sh) * sstride + ((i & msk) << 4));
+        }
+        continue;
+      }
       for (; i + 192 < elems; i += 256) {
         uint32_t i0 = i, i1 = i + 64, i2 = i + 128, i3 = i + 192;
         const uint4* s0 = reinterpret_cast<const uint4*>(
@@ -679,14 +715,13 @@ template <> __device__ __forceinline__ __hip_bfloat16 convert<__half, __hip_bflo
   return __float2bfloat16(__half2float(v));
 }
 
-// 8 elements per thread per iteration; 16B loads when SrcT is 2 bytes,
-// 32B (2x float4) when 4 bytes — always >= the 16B/lane coalescing sweet
-// spot on the wider side (guide G13).
-template <typename SrcT, typename DstT>
+// V elements per thread per iteration; 16B loads when SrcT is 2 bytes,
+// 32B (2x float4) when 4 bytes at V=8 — always >= the 16B/lane coalescing
+// sweet spot on the wider side (guide G13).  HIPSTORE_CAST_V sweeps V.
+template <typename SrcT, typename DstT, uint32_t V>
 __global__ void __launch_bounds__(256)
 cast_copy_kernel(const SrcT* __restrict__ src, DstT* __restrict__ dst,
                  uint64_t numel) {
-  constexpr uint32_t V = 8;
   uint64_t base = (uint64_t)(blockIdx.x * blockDim.x + threadIdx.x) * V;
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x * V;
   struct alignas(16) SrcVec { SrcT v[V]; };
@@ -710,13 +745,23 @@ cast_copy_kernel(const SrcT* __restrict__ src, DstT* __restrict__ dst,
 template <typename SrcT, typename DstT>
 static void launch_cast(uintptr_t src, uintptr_t dst, uint64_t numel,
                         hipStream_t stream) {
-  uint64_t work = (numel + 7) / 8;
+  const char* e = getenv("HIPSTORE_CAST_V");
+  uint32_t v = e ? (uint32_t)atoi(e) : 8u;
+  uint64_t work = (numel + v - 1) / v;
   uint32_t grid =
       (uint32_t)std::min<uint64_t>((work + 255) / 256, pick_cast_grid_cap());
   if (grid == 0) grid = 1;
-  hipLaunchKernelGGL((cast_copy_kernel<SrcT, DstT>), dim3(grid), dim3(256), 0,
-                     stream, reinterpret_cast<const SrcT*>(src),
-                     reinterpret_cast<DstT*>(dst), numel);
+  if (v >= 16) {
+    hipLaunchKernelGGL((cast_copy_kernel<SrcT, DstT, 16>), dim3(grid),
+                       dim3(256), 0, stream,
+                       reinterpret_cast<const SrcT*>(src),
+                       reinterpret_cast<DstT*>(dst), numel);
+  } else {
+    hipLaunchKernelGGL((cast_copy_kernel<SrcT, DstT, 8>), dim3(grid),
+                       dim3(256), 0, stream,
+                       reinterpret_cast<const SrcT*>(src),
+                       reinterpret_cast<DstT*>(dst), numel);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
