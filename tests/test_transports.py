@@ -189,3 +189,36 @@ def test_shm_put_segment_reuse_and_growth():
     (d4,) = buf.recv_handshake([Request(key="k")], "put", None)
     assert d4.seg_key != d3.seg_key
     volume_ctx.close()
+
+
+def test_delete_drops_volume_shm_segments():
+    """delete() must release the volume's SHM segments for the key (put-
+    and get-side), not just the index entry."""
+    import asyncio
+
+    import torch
+
+    from torchstore_amd.transport.base import TransportContext
+    from torchstore_amd.transport.shm import ShmTransportBuffer, ShmVolumeCache
+    from torchstore_amd.types import Request
+
+    ctx = TransportContext()
+    cache: ShmVolumeCache = ctx.cache(ShmVolumeCache)
+    # put-side segment via the handshake
+    buf = ShmTransportBuffer()
+    buf.client_uid = "c"
+    buf.alloc_sizes = [256]
+    buf.attach_volume(ctx)
+    buf.recv_handshake([Request(key="k1")], "put", None)
+    # get-side segment via volume_send
+    buf2 = ShmTransportBuffer()
+    buf2.client_uid = "c"
+    buf2.attach_volume(ctx)
+    asyncio.run(buf2.volume_send([Request(key="k1")], [torch.randn(8)]))
+    asyncio.run(buf2.volume_send([Request(key="k2")], [torch.randn(8)]))
+    assert len(cache.put_segments) == 1 and len(cache.get_segments) == 2
+    ctx.drop_key("k1")
+    assert not cache.put_segments
+    assert len(cache.get_segments) == 1
+    assert next(iter(cache.get_segments))[1] == "k2"
+    ctx.close()
