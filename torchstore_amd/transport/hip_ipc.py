@@ -157,9 +157,11 @@ class IpcOpenCache(TransportCache):
 class ChunkStagingCache(TransportCache):
     """Volume-side pool of <2 GiB staging chunks for the windowed path.
 
-    An operation acquires TWO chunks so the client can double-buffer:
-    window N+1's xGMI copy overlaps window N's commit RPC (the round-1
-    serial path lost ~25% bandwidth at 2 GiB to the per-window RPC).
+    An operation acquires THREE chunks: the client's chunk-reuse
+    distance then tolerates the volume's ASYNC commit copies (a commit
+    reply only guarantees the PREVIOUS window's copy), so client xGMI
+    copies, volume copies and commit RPCs all overlap.  Also owns the
+    direct push/pull per-op state (pending payloads / pinned exports).
     """
 
     def __init__(self):
