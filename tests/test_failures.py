@@ -132,3 +132,28 @@ async def test_repeated_lifecycle_no_leaks():
     # no stray actor children left behind
     leftovers = [p for p in multiprocessing.active_children()]
     assert leftovers == [], leftovers
+
+
+async def test_volume_stats_endpoint():
+    """Per-volume observability: entry/byte counts and tier occupancy."""
+    await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(),
+        storage_device="cpu",
+        storage_capacity_gb=1e-6,  # 1 kB primary -> second put spills
+    )
+    try:
+        await ts.put("a", torch.ones(128))          # 512 B -> primary
+        await ts.put("b", torch.ones(1024))         # 4 kB -> spilled
+        await ts.put("o", {"x": 1})
+        c = ts.client()
+        await c._ensure_volumes()
+        v = next(iter(c._volumes.values()))
+        stats = await v.handle.stats.call_one()
+        assert stats["entries"] == 3
+        assert stats["tensor_entries"] == 2 and stats["object_entries"] == 1
+        assert stats["tensor_bytes"] == 512 + 4096
+        assert stats["tier_primary_used"] == 512
+        assert stats["tier_capacity"] == 1000
+    finally:
+        await ts.shutdown()

@@ -405,5 +405,36 @@ class StorageVolume(Actor):
     def stored_keys(self) -> List[str]:
         return self.store.keys()
 
+    @endpoint
+    def stats(self) -> Dict[str, Any]:
+        """Per-volume storage observability: entry counts and resident
+        bytes by kind (plus tier occupancy for TieredStore volumes)."""
+        out: Dict[str, Any] = {
+            "volume_id": self.volume_id,
+            "device": str(self.device),
+            "entries": 0,
+            "tensor_entries": 0,
+            "shard_entries": 0,
+            "object_entries": 0,
+            "tensor_bytes": 0,
+        }
+        kv = getattr(self.store, "kv", {})
+        for entry in kv.values():
+            out["entries"] += 1
+            if isinstance(entry, _TensorEntry):
+                out["tensor_entries"] += 1
+                t = entry.tensor
+                out["tensor_bytes"] += t.numel() * t.element_size()
+            elif isinstance(entry, _ShardEntry):
+                out["shard_entries"] += 1
+                for _s, t in entry.shards.values():
+                    out["tensor_bytes"] += t.numel() * t.element_size()
+            else:
+                out["object_entries"] += 1
+        if isinstance(self.store, TieredStore):
+            out["tier_primary_used"] = self.store.primary_used
+            out["tier_capacity"] = self.store.capacity_bytes
+        return out
+
     def teardown_local(self):
         self.ctx.close()
