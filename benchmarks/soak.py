@@ -73,6 +73,19 @@ async def run(iters: int, report: int):
                 assert torch.equal(v, items[k]), k
             if i % 7 == 6:
                 await ts.delete_batch(list(items.keys()))
+            if dev == "cuda" and i % 50 == 49:
+                # periodic >=2 GiB traffic: auto-split + direct push/pull
+                # + epoch replacement + delete reclamation
+                huge = torch.empty(560_000_000, dtype=torch.float32,
+                                   device=dev)
+                huge.fill_(float(i))
+                await ts.put("s/huge", huge)
+                back = torch.zeros_like(huge)
+                await ts.get("s/huge", back)
+                assert back[::1_000_003].eq(float(i)).all()
+                del huge, back
+                if i % 100 == 99:
+                    await ts.delete("s/huge")
             if (i + 1) % report == 0:
                 torch.cuda.synchronize() if dev == "cuda" else None
                 used = (
