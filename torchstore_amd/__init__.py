@@ -48,6 +48,7 @@ __version__ = "0.1.0"
 
 __all__ = [
     "attach",
+    "init_logging",
     "client",
     "delete",
     "delete_batch",
