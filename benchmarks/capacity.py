@@ -21,11 +21,13 @@ import torchstore_amd as ts
 from torchstore_amd.strategy import LocalRankStrategy
 
 
-async def run(total_gb: float, chunk_mb: int, verify: bool):
+async def run(total_gb: float, chunk_mb: int, verify: bool,
+              capacity_gb: float = None):
     await ts.initialize(
         num_storage_volumes=1,
         strategy=LocalRankStrategy(),
         storage_device="auto",
+        storage_capacity_gb=capacity_gb,
     )
     try:
         chunk_bytes = chunk_mb << 20
@@ -151,11 +153,14 @@ def main():
                    help="wave-streamed fill: store holds --total-gb "
                         "resident (client frees sources between waves)")
     p.add_argument("--wave-gb", type=float, default=16.0)
+    p.add_argument("--capacity-gb", type=float, default=None,
+                   help="TieredStore primary capacity (overflow spills to host)")
     args = p.parse_args()
     if args.stream:
         asyncio.run(run_stream(args.total_gb, args.chunk_mb, args.wave_gb))
     else:
-        asyncio.run(run(args.total_gb, args.chunk_mb, not args.no_verify))
+        asyncio.run(run(args.total_gb, args.chunk_mb, not args.no_verify,
+                        args.capacity_gb))
 
 
 if __name__ == "__main__":
