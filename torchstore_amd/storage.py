@@ -273,6 +273,18 @@ class TieredStore(InMemoryStore):
             self.capacity_bytes,
         )
         if tensor.device != self.spill_device:
+            if self.spill_device.type == "cpu" and torch.cuda.is_available():
+                try:
+                    # PINNED spill: D2H lands at DMA rate and later reads
+                    # stage back through HBM at ~50 GB/s instead of the
+                    # ~13 GB/s pageable path
+                    pinned = torch.empty(
+                        tensor.shape, dtype=tensor.dtype, pin_memory=True
+                    )
+                    pinned.copy_(tensor)
+                    return pinned
+                except RuntimeError:  # pinned allocation failed: fall back
+                    pass
             tensor = tensor.to(self.spill_device)
         return tensor
 

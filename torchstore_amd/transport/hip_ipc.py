@@ -1013,6 +1013,16 @@ class HipIpcTransportBuffer(TransportBuffer):
             if kind == "fetch_obj" or not isinstance(v, torch.Tensor):
                 reply.append(("inline", v))
                 continue
+            if v.device.type != "cuda" and kind in ("ipc", "bounce"):
+                # tier-spilled (host-resident) value headed for a GPU dest:
+                # stage through HBM and take the one-sided path — the
+                # inline-RPC serialization ran at ~2 GB/s, a pinned H2D +
+                # device copy at ~40 GB/s
+                if torch.cuda.is_available():
+                    v = v.to(
+                        torch.device("cuda", torch.cuda.current_device()),
+                        non_blocking=False,
+                    )
             if kind == "fetch_inline" or v.device.type != "cuda":
                 reply.append(("inline", v))
                 continue
