@@ -515,6 +515,14 @@ class HipIpcTransportBuffer(TransportBuffer):
             launched = False
             for meta in metas:
                 value = volume.store.fetch(meta)
+                if (
+                    isinstance(value, torch.Tensor)
+                    and value.device.type == "cpu"
+                    and device.type == "cuda"
+                ):
+                    # tier-spilled value: stage through HBM so the client
+                    # pulls at device rate instead of host-memory rate
+                    value = value.to(device)
                 if not isinstance(value, torch.Tensor) or value.device != device:
                     stash.append(None)
                     descs.append(None)
