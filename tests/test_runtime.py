@@ -160,3 +160,30 @@ async def test_giant_rpc_frame_over_2gib():
     finally:
         await mesh.stop()
         await close_connections()
+
+
+async def test_unserializable_argument_does_not_leak_pending():
+    """A call whose ARGUMENTS fail to serialize must raise cleanly and
+    leave no pending-future entry behind (it would leak for the life of
+    the connection otherwise)."""
+    mesh = spawn_actors(1, Echo, "serde-err")
+    try:
+        h = mesh.handles[0]
+        assert await h.add.call_one(1, 2) == 3
+        from torchstore_amd.runtime.actor import get_connection
+
+        conn = await get_connection(h.host, h.port)
+        before = len(conn._pending)
+
+        class Unpicklable:
+            def __reduce__(self):
+                raise TypeError("nope")
+
+        with pytest.raises(Exception):
+            await h.add.call_one(Unpicklable(), 2)
+        assert len(conn._pending) == before
+        # connection still healthy for further calls
+        assert await h.add.call_one(10, 5) == 15
+    finally:
+        await mesh.stop()
+        await close_connections()

@@ -230,10 +230,16 @@ class RpcConnection:
         req_id = next(self._ids)
         fut: asyncio.Future = asyncio.get_running_loop().create_future()
         self._pending[req_id] = fut
-        await _write_frame(
-            self._writer, self._wlock, KIND_REQUEST, req_id,
-            (method, args, kwargs),
-        )
+        try:
+            await _write_frame(
+                self._writer, self._wlock, KIND_REQUEST, req_id,
+                (method, args, kwargs),
+            )
+        except BaseException:
+            # the request never left (serde error on args, closed socket):
+            # drop the pending entry or it leaks for the connection's life
+            self._pending.pop(req_id, None)
+            raise
         return await fut
 
     async def close(self) -> None:
