@@ -753,3 +753,63 @@ async def test_huge_tensor_autosplit_region_fetch():
         assert full[::997][:, 1].eq(2.0).all()
 
     await _with_store(body, transport=TransportType.HIP_IPC)
+
+
+class GpuHammer(Actor):
+    """Concurrent GPU client: interleaved IPC puts/gets of its keyspace."""
+
+    def __init__(self, controller):
+        import os
+
+        from torchstore_amd.runtime import actor_context
+
+        self.rank = actor_context().rank
+        os.environ["RANK"] = str(self.rank)
+        torch.cuda.set_device(0)
+        ts.attach(controller, SingletonStrategy(transport=TransportType.HIP_IPC))
+
+    @endpoint
+    async def run(self, rounds: int):
+        for i in range(rounds):
+            stamp = float(self.rank * 1000 + i)
+            t = torch.full((256, 257), stamp, device="cuda")
+            await ts.put(f"g{self.rank}/k{i % 3}", t)
+            out = await ts.get(f"g{self.rank}/k{i % 3}")
+            torch.cuda.synchronize()
+            u = out.unique()
+            if u.numel() != 1 or u.item() != stamp:
+                return f"rank {self.rank} torn read at {i}: {u.tolist()[:4]}"
+            # shared key: concurrent writers, value must be self-consistent
+            await ts.put("g/shared", torch.full((64,), stamp, device="cuda"))
+            got = await ts.get("g/shared")
+            torch.cuda.synchronize()
+            if got.unique().numel() != 1:
+                return f"rank {self.rank} torn shared read at {i}"
+        return "ok"
+
+
+@requires_gpu
+async def test_concurrent_gpu_clients_over_ipc():
+    """Two client PROCESSES drive one GPU volume over real HIP IPC
+    concurrently: stream-pool sharing, export/open caches and in-place
+    overwrites under true cross-process contention."""
+    from torchstore_amd.runtime import spawn_actors
+
+    controller = await ts.initialize(
+        num_storage_volumes=1,
+        strategy=SingletonStrategy(),
+        storage_device="auto",
+    )
+    mesh = None
+    try:
+        import asyncio as aio
+
+        mesh = await aio.to_thread(
+            spawn_actors, 2, GpuHammer, "ghammer", controller
+        )
+        res = await mesh.run.call(40)
+        assert res == ["ok", "ok"], res
+    finally:
+        if mesh is not None:
+            await mesh.stop()
+        await ts.shutdown()
