@@ -187,3 +187,29 @@ async def test_unserializable_argument_does_not_leak_pending():
     finally:
         await mesh.stop()
         await close_connections()
+
+
+class SlowActor(Actor):
+    @endpoint
+    async def nap(self, seconds: float):
+        await asyncio.sleep(seconds)
+        return "awake"
+
+
+async def test_rpc_timeout_opt_in(monkeypatch):
+    """TORCHSTORE_AMD_RPC_TIMEOUT turns a hung call into TimeoutError;
+    unset (default) calls wait indefinitely."""
+    mesh = spawn_actors(1, SlowActor, "slow")
+    try:
+        h = mesh.handles[0]
+        assert await h.nap.call_one(0.01) == "awake"
+        monkeypatch.setenv("TORCHSTORE_AMD_RPC_TIMEOUT", "0.2")
+        with pytest.raises(asyncio.TimeoutError):
+            await h.nap.call_one(5.0)
+        monkeypatch.delenv("TORCHSTORE_AMD_RPC_TIMEOUT")
+        # connection still serves later calls (the slow one finishes
+        # server-side and its orphaned reply is dropped)
+        assert await h.nap.call_one(0.01) == "awake"
+    finally:
+        await mesh.stop()
+        await close_connections()

@@ -80,6 +80,21 @@ async def close_connections() -> None:
 # ---------------------------------------------------------------------------
 
 
+def _rpc_timeout() -> float:
+    """Opt-in per-call RPC timeout (seconds; 0 = wait forever).
+
+    Off by default: windowed multi-GB transfers legitimately hold RPCs
+    open for seconds.  Production deployments that prefer failing fast on
+    a hung volume set TORCHSTORE_AMD_RPC_TIMEOUT comfortably above their
+    largest transfer (e.g. 120)."""
+    import os
+
+    try:
+        return float(os.environ.get("TORCHSTORE_AMD_RPC_TIMEOUT", "0") or 0)
+    except ValueError:
+        return 0.0
+
+
 class _Endpoint:
     __slots__ = ("_handle", "_name")
 
@@ -89,6 +104,11 @@ class _Endpoint:
 
     async def call_one(self, *args, **kwargs) -> Any:
         conn = await get_connection(self._handle.host, self._handle.port)
+        t = _rpc_timeout()
+        if t > 0:
+            return await asyncio.wait_for(
+                conn.call(self._name, *args, **kwargs), timeout=t
+            )
         return await conn.call(self._name, *args, **kwargs)
 
     # alias so a single handle can stand in where mesh semantics are expected
