@@ -150,6 +150,9 @@ async def test_volume_stats_endpoint():
         await c._ensure_volumes()
         v = next(iter(c._volumes.values()))
         stats = await v.handle.stats.call_one()
+        agg = await ts.stats()
+        assert agg["controller"]["keys"] == 3
+        assert agg["volumes"][0]["entries"] == 3
         assert stats["entries"] == 3
         assert stats["tensor_entries"] == 2 and stats["object_entries"] == 1
         assert stats["tensor_bytes"] == 512 + 4096

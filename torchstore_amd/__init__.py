@@ -30,6 +30,7 @@ from torchstore_amd.api import (
     put_state_dict,
     reset_client,
     shutdown,
+    stats,
 )
 from torchstore_amd.spmd import SPMDEnv, initialize_spmd, shutdown_spmd
 from torchstore_amd.strategy import (
@@ -63,6 +64,7 @@ __all__ = [
     "put_state_dict",
     "reset_client",
     "shutdown",
+    "stats",
     "SPMDEnv",
     "initialize_spmd",
     "shutdown_spmd",

@@ -204,6 +204,18 @@ async def exists(key: str, store_name: str = DEFAULT_STORE) -> bool:
     return await client(store_name).exists(key)
 
 
+async def stats(store_name: str = DEFAULT_STORE) -> Dict[str, Any]:
+    """Whole-store observability: the controller's index stats plus every
+    volume's entry/byte/tier counters (beyond reference parity)."""
+    c = client(store_name)
+    volumes = await c._ensure_volumes()
+    ctrl, *vols = await asyncio.gather(
+        c._controller.stats.call_one(),
+        *(v.handle.stats.call_one() for v in volumes.values()),
+    )
+    return {"controller": ctrl, "volumes": list(vols)}
+
+
 async def put_state_dict(
     state_dict: Dict[str, Any], key: str, store_name: str = DEFAULT_STORE, **kw
 ) -> None:
