@@ -100,7 +100,9 @@ async def put_state_dict(
     direct: bool = False,
     rank: Optional[int] = None,
     world_size: Optional[int] = None,
+    direct_rdma: bool = False,  # reference-compat alias for ``direct``
 ) -> None:
+    direct = direct or direct_rdma
     if direct:
         from torchstore_amd.weight_sync import DirectWeightSyncSource
 
@@ -135,7 +137,9 @@ async def get_state_dict(
     user_state_dict: Optional[Dict[str, Any]] = None,
     strict: bool = True,
     direct: bool = False,
+    direct_rdma: bool = False,  # reference-compat alias for ``direct``
 ) -> Dict[str, Any]:
+    direct = direct or direct_rdma
     if direct:
         from torchstore_amd.weight_sync import DirectWeightSyncDest
 
