@@ -303,3 +303,17 @@ async def test_plan_cache_dest_identity(fake_codec):
         assert fake_codec.read_count == n_reads + 2
 
     await _with_store(body)
+
+
+async def test_direct_rdma_alias(fake_codec):
+    """Reference-compat kwarg: direct_rdma=True behaves as direct=True."""
+
+    async def body():
+        w = torch.randn(8, 8)
+        await ts.put_state_dict({"w": w}, "alias", direct_rdma=True,
+                                rank=0, world_size=1)
+        dest = {"w": torch.zeros(8, 8)}
+        out = await ts.get_state_dict("alias", dest, direct_rdma=True)
+        assert torch.equal(out["w"], w)
+
+    await _with_store(body)
