@@ -222,3 +222,105 @@ def test_delete_drops_volume_shm_segments():
     assert len(cache.get_segments) == 1
     assert next(iter(cache.get_segments))[1] == "k2"
     ctx.close()
+
+
+def test_shm_adopted_entry_zero_copy_warm_get():
+    """A CPU volume ADOPTS the typed view of its own put segment as
+    storage; warm gets of that key return the SAME segment descriptor
+    with no volume-side copy (reference warm-get reuse,
+    shared_memory.py:340-360 + our ownership inversion)."""
+    import asyncio
+
+    import torch
+
+    from torchstore_amd.transport.base import TransportContext
+    from torchstore_amd.transport.shm import (
+        ShmTransportBuffer,
+        ShmVolumeCache,
+        _typed_view,
+    )
+    from torchstore_amd.types import Request
+
+    ctx = TransportContext()
+    cache: ShmVolumeCache = ctx.cache(ShmVolumeCache)
+    # put handshake allocates the segment; the volume adopts the view
+    buf = ShmTransportBuffer()
+    buf.client_uid = "c"
+    buf.alloc_sizes = [1024]
+    buf.attach_volume(ctx)
+    (desc,) = buf.recv_handshake([Request(key="k")], "put", None)
+    seg = cache.put_segments[("c", "k")][1]
+    stored = _typed_view(seg, desc.with_layout(torch.float32, (256,)))
+    stored.copy_(torch.arange(256, dtype=torch.float32))
+
+    # warm get: the reply descriptor POINTS AT the adopted segment —
+    # same seg_key, zero copies (desc_by_storage fast path)
+    getbuf = ShmTransportBuffer()
+    getbuf.client_uid = "reader"
+    getbuf.attach_volume(ctx)
+    reply = asyncio.run(getbuf.volume_send([Request(key="k")], [stored]))
+    kind, rdesc = reply[0]
+    assert kind == "seg"
+    assert rdesc.seg_key == desc.seg_key, "warm get must reuse the segment"
+    assert rdesc.dtype == torch.float32 and rdesc.shape == (256,)
+    assert not cache.get_segments, "no response segment should be allocated"
+    ctx.close()
+
+
+def test_shm_descriptor_layout_roundtrip():
+    """with_layout carries dtype/shape; _typed_view reconstructs exactly
+    (incl. empty and odd shapes) from the raw byte segment."""
+    import torch
+
+    from torchstore_amd.transport.shm import (
+        _allocate_segment,
+        _segment_descriptor,
+        _typed_view,
+    )
+
+    seg = _allocate_segment(4096)
+    base = _segment_descriptor(seg)
+    for dtype, shape in [
+        (torch.float32, (16, 16)),
+        (torch.bfloat16, (3, 5, 7)),
+        (torch.int64, (512,)),
+        (torch.float16, (0,)),
+        (torch.uint8, ()),
+    ]:
+        d = base.with_layout(dtype, shape)
+        v = _typed_view(seg, d)
+        assert v.dtype == dtype and tuple(v.shape) == tuple(shape)
+        if v.numel():
+            v.fill_(1 if dtype == torch.int64 else 1.0)
+            v2 = _typed_view(seg, d)
+            assert v2.reshape(-1)[0].item() == 1
+
+
+def test_shm_unpin_on_close(monkeypatch):
+    """ctx.close() unregisters every pinned segment exactly once."""
+    import torch
+
+    from torchstore_amd.transport import shm
+
+    calls = {"reg": 0, "unreg": 0}
+
+    class FakeCudart:
+        def cudaHostRegister(self, ptr, nbytes, flags):
+            calls["reg"] += 1
+            return 0
+
+        def cudaHostUnregister(self, ptr):
+            calls["unreg"] += 1
+            return 0
+
+    monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(torch.cuda, "cudart", lambda: FakeCudart())
+    pinned = {}
+    s1 = shm._allocate_segment(1024)
+    s2 = shm._allocate_segment(1024)
+    shm._try_pin(s1, pinned)
+    shm._try_pin(s1, pinned)  # idempotent per segment
+    shm._try_pin(s2, pinned)
+    assert calls["reg"] == 2 and len(pinned) == 2
+    shm._unpin_all(pinned)
+    assert calls["unreg"] == 2 and not pinned
