@@ -51,6 +51,10 @@ class Rank(Actor):
         row_sharded = distribute_tensor(full, mesh, [Shard(0)])
         await api.put("w", row_sharded)
 
+        # the commit gate makes "w" readable only once EVERY coordinate has
+        # stored its shard — barrier before reading, like any SPMD step
+        dist.barrier()
+
         col_dest = distribute_tensor(torch.zeros_like(full), mesh, [Shard(1)])
         out = await api.get("w", col_dest)
 
