@@ -74,28 +74,51 @@ class SdHammer(Actor):
     """Concurrent state_dict pushers/readers under distinct keys plus a
     reader racing a writer on one key (commit-marker consistency)."""
 
-    def __init__(self, controller):
+    def __init__(self, controller, transport=None):
+        from torchstore_amd.transport import TransportType
+
         self.rank = actor_context().rank
-        api.attach(controller, SingletonStrategy())
+        api.attach(
+            controller,
+            SingletonStrategy(
+                transport=TransportType(transport) if transport else None
+            ),
+        )
 
     @endpoint
     async def run(self, rounds: int):
-        from torchstore_amd import state_dict as sdmod
-
         for i in range(rounds):
+            # >8 entries of UNLIKE sizes: the pipeline splits each op into
+            # concurrent sub-batches over ONE cached transport pair, so a
+            # send pairing with the wrong recv would corrupt or hang
             sd = {
                 "m": {
-                    "w": torch.full((128, 8), float(self.rank * 100 + i)),
-                    "b": torch.full((8,), float(i)),
+                    f"l{j}": torch.full(
+                        (16 + 16 * j, 8), float(self.rank * 100 + i)
+                    )
+                    for j in range(12)
                 },
                 "step": i,
             }
-            await api.put_state_dict(sd, f"sd{self.rank}")
-            out = await api.get_state_dict(f"sd{self.rank}")
-            if out["step"] != i or not out["m"]["w"].eq(
-                float(self.rank * 100 + i)
-            ).all():
-                raise AssertionError("own state_dict readback wrong")
+            # two CONCURRENT ops on the same pair
+            await asyncio.gather(
+                api.put_state_dict(sd, f"sd{self.rank}"),
+                api.put_state_dict(sd, f"sd{self.rank}x"),
+            )
+            out, _outx = await asyncio.gather(
+                api.get_state_dict(f"sd{self.rank}"),
+                api.get_state_dict(f"sd{self.rank}x"),
+            )
+            if out["step"] != i:
+                raise AssertionError("own state_dict step wrong")
+            for j in range(12):
+                w = out["m"][f"l{j}"]
+                if tuple(w.shape) != (16 + 16 * j, 8) or not w.eq(
+                    float(self.rank * 100 + i)
+                ).all():
+                    raise AssertionError(
+                        f"own state_dict readback wrong at l{j}"
+                    )
             # race the OTHER rank's key: either absent (no push yet) or a
             # complete, self-consistent snapshot — never a torn one
             peer = 1 - self.rank
@@ -103,13 +126,20 @@ class SdHammer(Actor):
                 got = await api.get_state_dict(f"sd{peer}")
             except (RuntimeError, KeyError):
                 continue  # no commit marker yet — acceptable
-            u = got["m"]["w"].unique()
+            u = got["m"]["l0"].unique()
             if u.numel() != 1:
                 raise AssertionError(f"torn peer state_dict: {u}")
         return "ok"
 
 
-async def test_concurrent_state_dict_exchange():
+import pytest
+
+
+@pytest.mark.parametrize("transport", [None, "gloo"])
+async def test_concurrent_state_dict_exchange(transport):
+    """transport=gloo drives the pair-PG path with up to 8 concurrent
+    sub-batches per op over ONE cached PG: the per-op TAGS must pair
+    every send with its own recv (tag collisions)."""
     controller = await ts.initialize(
         num_storage_volumes=1,
         strategy=SingletonStrategy(),
@@ -117,7 +147,9 @@ async def test_concurrent_state_dict_exchange():
     )
     mesh = None
     try:
-        mesh = await asyncio.to_thread(spawn_actors, 2, SdHammer, "sdh", controller)
+        mesh = await asyncio.to_thread(
+            spawn_actors, 2, SdHammer, "sdh", controller, transport
+        )
         res = await mesh.run.call(6)
         assert res == ["ok", "ok"]
     finally:
