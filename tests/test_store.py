@@ -369,3 +369,18 @@ async def test_noncontiguous_put(transport):
             assert torch.equal(dest, v), k
 
     await _with_store(transport, body)
+
+
+async def test_get_batch_all_or_nothing(transport):
+    """A missing key fails the whole get_batch (reference locate raises
+    before any transport work begins — no partial results)."""
+
+    async def body():
+        await ts.put_batch({"ab/x": torch.ones(4), "ab/y": torch.ones(4)})
+        with pytest.raises(KeyError, match="does not exist"):
+            await ts.get_batch({"ab/x": None, "ab/missing": None})
+        # the present keys remain readable afterwards
+        out = await ts.get_batch({"ab/x": None, "ab/y": None})
+        assert out["ab/x"].eq(1).all() and out["ab/y"].eq(1).all()
+
+    await _with_store(transport, body)
