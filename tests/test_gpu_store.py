@@ -813,3 +813,43 @@ async def test_concurrent_gpu_clients_over_ipc():
         if mesh is not None:
             await mesh.stop()
         await ts.shutdown()
+
+
+@requires_gpu
+async def test_windowed_fallback_oversized_shard():
+    """An OVERSIZED user shard (tensor_slice set → no auto-split, and a
+    ≥2 GiB volume payload → push/pull cannot apply) exercises the 3-deep
+    pipelined WINDOWED path on hardware in both directions."""
+    from torchstore_amd.types import LocalShard, TensorSlice
+
+    async def body():
+        rows = 640_000_000  # 2.56 GB f32, single shard of a 2-shard mesh
+        t = torch.empty(rows, device="cuda")
+        t[::1_000_003] = torch.arange(
+            (rows + 1_000_002) // 1_000_003, device="cuda", dtype=torch.float32
+        )
+        sl = TensorSlice(
+            offsets=(0,), local_shape=(rows,), global_shape=(2 * rows,),
+            coordinates=(0,), mesh_shape=(2,),
+        )
+        sl2 = TensorSlice(
+            offsets=(rows,), local_shape=(rows,), global_shape=(2 * rows,),
+            coordinates=(1,), mesh_shape=(2,),
+        )
+        await ts.put("win/w", LocalShard(tensor=t, slice=sl))  # windowed put
+        # commit the gate with a second (small-pattern) shard
+        t2 = torch.ones(rows, device="cuda")
+        await ts.put("win/w", LocalShard(tensor=t2, slice=sl2))
+        dest = LocalShard(tensor=torch.zeros(rows, device="cuda"), slice=sl)
+        await ts.get("win/w", dest)  # windowed get (pull can't export 2.5GB)
+        torch.cuda.synchronize()
+        assert torch.equal(dest.tensor[::1_000_003], t[::1_000_003])
+        assert dest.tensor[1].item() == 0.0  # untouched positions intact
+        # in-place windowed RE-put reuses the stored payload
+        t[::1_000_003] += 1.0
+        await ts.put("win/w", LocalShard(tensor=t, slice=sl))
+        await ts.get("win/w", dest)
+        torch.cuda.synchronize()
+        assert torch.equal(dest.tensor[::1_000_003], t[::1_000_003])
+
+    await _with_store(body, transport=TransportType.HIP_IPC)
