@@ -86,3 +86,27 @@ def test_expected_coordinates_cartesian():
     assert len(coords) == 6
     assert tuple(coords[0]) == (0, 0) and tuple(coords[-1]) == (1, 2)
     assert list(TensorSlice.expected_coordinates(())) == [()]
+
+
+def test_plan_row_split_math():
+    from torchstore_amd.client import _plan_row_split
+
+    G = 1 << 30
+    # under the limit: no split
+    assert _plan_row_split((1024, 1024), 4) is None
+    # 2.5 GB f32 1-D: pieces of 1 GiB rows
+    rows = 640_000_000
+    rpp, k = _plan_row_split((rows,), 4)
+    assert rpp == G // 4 and k == 3
+    assert (k - 1) * rpp < rows <= k * rpp
+    # 2-D: row granularity respected
+    rpp, k = _plan_row_split((300_000_000, 2), 4)
+    assert rpp == G // 8 and k == 3
+    # single-row giant: unsplittable
+    assert _plan_row_split((1, 1 << 30), 4) is None
+    # a row itself over the limit: unsplittable
+    assert _plan_row_split((4, 1 << 30), 4) is None
+    # 0-d / empty
+    assert _plan_row_split((), 4) is None
+    # exactly at the limit splits
+    assert _plan_row_split(((1 << 31) // 4, 1), 4) is not None

@@ -63,6 +63,32 @@ class _CachedFetchPlan:
     busy: bool = False  # a concurrent identical call must not share subs
 
 
+def _plan_row_split(
+    shape: Tuple[int, ...], elem_size: int, limit: int = 1 << 31,
+    piece_bytes: int = 1 << 30,
+) -> Optional[Tuple[int, int]]:
+    """(rows_per_piece, n_pieces) for splitting an oversized tensor along
+    dim 0, or None when splitting is impossible/unneeded (pure math —
+    unit-tested on CPU)."""
+    if not shape:
+        return None
+    numel = 1
+    for d in shape:
+        numel *= d
+    nbytes = numel * elem_size
+    if nbytes < limit:
+        return None
+    rows = shape[0]
+    row_bytes = nbytes // max(rows, 1)
+    if rows < 2 or row_bytes == 0 or row_bytes >= limit:
+        return None
+    rows_per_piece = max(1, piece_bytes // row_bytes)
+    k = (rows + rows_per_piece - 1) // rows_per_piece
+    if k < 2:
+        return None
+    return rows_per_piece, k
+
+
 def _split_huge_tensors(
     requests: List[Request], ref: StorageVolumeRef
 ) -> List[Request]:
@@ -96,16 +122,14 @@ def _split_huge_tensors(
             out.append(r)
             continue
         tc = t.contiguous()
-        rows = tc.shape[0]
-        row_bytes = tc.numel() * tc.element_size() // max(rows, 1)
-        if rows < 2 or row_bytes == 0 or row_bytes >= IPC_BLOCK_LIMIT:
+        plan = _plan_row_split(
+            tuple(tc.shape), tc.element_size(), IPC_BLOCK_LIMIT
+        )
+        if plan is None:
             out.append(r)  # single-row giants keep the windowed path
             continue
-        rows_per_piece = max(1, (1 << 30) // row_bytes)
-        k = (rows + rows_per_piece - 1) // rows_per_piece
-        if k < 2:
-            out.append(r)
-            continue
+        rows_per_piece, k = plan
+        rows = tc.shape[0]
         gshape = tuple(tc.shape)
         for j in range(k):
             r0 = j * rows_per_piece
