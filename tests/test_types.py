@@ -110,3 +110,47 @@ def test_plan_row_split_math():
     assert _plan_row_split((), 4) is None
     # exactly at the limit splits
     assert _plan_row_split(((1 << 31) // 4, 1), 4) is not None
+
+
+def test_strategy_selection_units():
+    """Placement-strategy mapping units (reference strategy.py:111-188)."""
+    import os
+
+    from torchstore_amd.strategy import (
+        HostStrategy,
+        LocalRankStrategy,
+        SingletonStrategy,
+        strategy_from_spec,
+    )
+    from torchstore_amd.transport import TransportType
+
+    lr = LocalRankStrategy()
+    os.environ["RANK"] = "3"
+    try:
+        assert lr.client_id() == "3"
+        assert lr.select_volume_id(["0", "1", "2", "3"]) == "3"
+        # more clients than volumes: deterministic modulo placement
+        assert lr.select_volume_id(["0", "1"]) == "1"
+        assert lr.num_volumes_for(8, 2) == 8
+    finally:
+        del os.environ["RANK"]
+
+    hs = HostStrategy()
+    os.environ["HOSTNAME"] = "hx"
+    try:
+        assert hs.select_volume_id(["ha", "hx"]) == "hx"
+        assert hs.select_volume_id(["ha", "hb"]) == "ha"  # fallback sorted
+        assert hs.num_volumes_for(16, 2) == 2
+    finally:
+        del os.environ["HOSTNAME"]
+
+    ss = SingletonStrategy(transport=TransportType.RPC)
+    assert ss.select_volume_id(["b", "a"]) == "a"
+    assert ss.num_volumes_for(99, 9) == 1
+
+    # spec round-trips the kind AND the forced transport (attaching
+    # clients reconstruct the spawning strategy from it)
+    back = strategy_from_spec(ss.spec())
+    assert type(back) is SingletonStrategy
+    assert back.transport == TransportType.RPC
+    assert type(strategy_from_spec(None)) is SingletonStrategy
