@@ -154,3 +154,48 @@ def test_strategy_selection_units():
     assert type(back) is SingletonStrategy
     assert back.transport == TransportType.RPC
     assert type(strategy_from_spec(None)) is SingletonStrategy
+
+
+def test_gpu_descriptor_builders():
+    """CPU units for the kernel-descriptor builders: shapes/strides in
+    BYTES, contiguous fast paths, rejection of inexpressible layouts."""
+    from torchstore_amd.ops.gpu import (
+        _desc_view_to_ptr,
+        _pitched_params,
+        _slice_desc,
+    )
+
+    a = torch.zeros(8, 6)
+    b = torch.zeros(8, 6)
+    # both contiguous → flat byte descriptor
+    d = _slice_desc(a, b)
+    assert d == (a.data_ptr(), b.data_ptr(), 8 * 6 * 4, [], [], [])
+    # strided src → rows of row_bytes with byte strides
+    big = torch.zeros(8, 12)
+    src = big[:, 3:9]
+    d = _slice_desc(src, b)
+    assert d[2] == 6 * 4 and d[3] == [8]
+    assert d[4] == [12 * 4] and d[5] == [6 * 4]
+    # shape/dtype mismatch → None (caller falls back)
+    assert _slice_desc(a, torch.zeros(6, 8)) is None
+    assert _slice_desc(a, torch.zeros(8, 6, dtype=torch.float16)) is None
+    # inner-stride != 1 → None
+    assert _slice_desc(a.t(), torch.zeros(6, 8)) is None
+    # empty → () sentinel (skip, not fallback)
+    assert _slice_desc(torch.zeros(0), torch.zeros(0)) == ()
+    # 0-d → single-element descriptor
+    d = _slice_desc(torch.zeros(()), torch.zeros(()))
+    assert d[2] == 4 and d[3] == []
+
+    # fused view→ptr: contiguous dest strides synthesized for src's shape
+    v = big[2:6, 3:9]  # 4x6 strided view
+    d = _desc_view_to_ptr(v, 0xDEAD)
+    assert d[1] == 0xDEAD and d[2] == 6 * 4
+    assert d[3] == [4] and d[4] == [12 * 4] and d[5] == [6 * 4]
+    assert _desc_view_to_ptr(a.t(), 0) is None
+
+    # pitched params: (pitch, width, height) in bytes
+    assert _pitched_params(v) == (12 * 4, 6 * 4, 4)
+    assert _pitched_params(torch.zeros(10)) == (40, 40, 1)
+    assert _pitched_params(a.t()) is None
+    assert _pitched_params(torch.zeros(2, 3, 4)) is None
